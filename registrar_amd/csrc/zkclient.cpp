@@ -1,0 +1,911 @@
+// zkclient.cpp — native ZooKeeper client implementation (see zkclient.hpp).
+#include "zkclient.hpp"
+
+#include <arpa/inet.h>
+#include <fcntl.h>
+#include <netdb.h>
+#include <netinet/in.h>
+#include <netinet/tcp.h>
+#include <sys/socket.h>
+#include <unistd.h>
+
+#include <algorithm>
+#include <chrono>
+#include <future>
+#include <thread>
+
+namespace registrar {
+namespace zk {
+
+const char* session_event_name(SessionEvent::Type t) {
+  switch (t) {
+    case SessionEvent::Type::Connected:
+      return "connect";
+    case SessionEvent::Type::Disconnected:
+      return "close";
+    case SessionEvent::Type::Expired:
+      return "session_expired";
+    case SessionEvent::Type::ConnectAttempt:
+      return "attempt";
+    case SessionEvent::Type::Closed:
+      return "closed";
+  }
+  return "?";
+}
+
+namespace {
+void set_nonblock(int fd) {
+  int fl = fcntl(fd, F_GETFL, 0);
+  fcntl(fd, F_SETFL, fl | O_NONBLOCK);
+}
+void set_nodelay(int fd) {
+  int one = 1;
+  setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+}
+}  // namespace
+
+struct ZkClient::Impl {
+  enum class Phase { Idle, TcpConnecting, Handshaking, Ready, Stopped };
+
+  struct Pending {
+    int32_t xid;
+    int32_t op;
+    // rc + body reader (null unless rc==kZOk and the op has a body)
+    std::function<void(int rc, JuteReader* r)> done;
+  };
+
+  ZkClientConfig cfg;
+  Logger log;
+  EventLoop loop;
+  std::thread thread;
+  std::atomic<bool> started{false};
+  std::atomic<bool> closed{false};
+
+  // ---- loop-thread-only state ----
+  Phase phase = Phase::Idle;
+  int fd = -1;
+  std::string inbuf;
+  size_t inpos = 0;
+  std::string outbuf;
+  bool flush_scheduled = false;
+  std::deque<Pending> pending;
+  int32_t next_xid = 1;
+  size_t server_idx = 0;
+  Backoff connect_backoff;    // pre-session (1 s → 90 s, infinite by default)
+  Backoff reconnect_backoff;  // session-preserving (10 ms → 1 s)
+  bool have_session = false;
+  int64_t sid = 0;
+  std::string passwd;
+  int negotiated_timeout = 0;
+  int64_t last_zxid = 0;
+  int64_t last_recv = 0;
+  EventLoop::TimerId conn_timer = 0;
+  EventLoop::TimerId ping_timer = 0;
+  EventLoop::TimerId retry_timer = 0;
+  std::string current_server;
+
+  // ---- shared state ----
+  std::atomic<SessionState> state{SessionState::Connecting};
+  std::atomic<int64_t> session_id_pub{0};
+  std::atomic<int64_t> negotiated_pub{0};
+  std::mutex ev_mu;
+  std::condition_variable ev_cv;
+  std::vector<SessionEvent> ev_queue;
+  EventCallback ev_cb;
+  WatchCallback watch_cb;
+  bool settled = false;  // first connect resolved (ok or fail)
+
+  Impl(ZkClientConfig c, Logger l) : cfg(std::move(c)), log(l.child("zookeeper")) {
+    log.set_level(cfg.log_level);
+    connect_backoff.initial_ms = cfg.connect_initial_delay_ms;
+    connect_backoff.max_ms = cfg.connect_max_delay_ms;
+    connect_backoff.max_attempts = cfg.connect_max_attempts;
+    reconnect_backoff.initial_ms = cfg.reconnect_initial_delay_ms;
+    reconnect_backoff.max_ms = cfg.reconnect_max_delay_ms;
+    reconnect_backoff.max_attempts = -1;
+  }
+
+  // ---------------- events ----------------
+
+  void emit(SessionEvent ev) {
+    {
+      std::lock_guard<std::mutex> g(ev_mu);
+      ev_queue.push_back(ev);
+      if (ev.type != SessionEvent::Type::ConnectAttempt) settled_check(ev);
+    }
+    ev_cv.notify_all();
+    if (ev_cb) ev_cb(ev);
+  }
+
+  // ev_mu held
+  void settled_check(const SessionEvent& ev) {
+    if (ev.type == SessionEvent::Type::Connected || ev.type == SessionEvent::Type::Expired ||
+        ev.type == SessionEvent::Type::Closed)
+      settled = true;
+  }
+
+  // ---------------- connect machinery (loop thread) ----------------
+
+  void start_connect() {
+    if (phase == Phase::Stopped || closed.load()) return;
+    const ServerAddr& srv = cfg.servers[server_idx % cfg.servers.size()];
+    server_idx++;
+    current_server = srv.host + ":" + std::to_string(srv.port);
+    log.debug("connecting", {{"server", Json(current_server)}});
+
+    struct sockaddr_in addr;
+    memset(&addr, 0, sizeof(addr));
+    addr.sin_family = AF_INET;
+    addr.sin_port = htons(static_cast<uint16_t>(srv.port));
+    if (inet_pton(AF_INET, srv.host.c_str(), &addr.sin_addr) != 1) {
+      struct hostent* he = gethostbyname(srv.host.c_str());
+      if (!he || he->h_addrtype != AF_INET) {
+        on_connect_failed("resolve failed");
+        return;
+      }
+      memcpy(&addr.sin_addr, he->h_addr_list[0], sizeof(addr.sin_addr));
+    }
+
+    fd = socket(AF_INET, SOCK_STREAM | SOCK_CLOEXEC, 0);
+    if (fd < 0) {
+      on_connect_failed("socket failed");
+      return;
+    }
+    set_nonblock(fd);
+    set_nodelay(fd);
+    int rc = ::connect(fd, reinterpret_cast<struct sockaddr*>(&addr), sizeof(addr));
+    phase = Phase::TcpConnecting;
+    conn_timer = loop.schedule(cfg.connect_timeout_ms, [this] {
+      conn_timer = 0;
+      on_connect_failed("connect timeout");
+    });
+    if (rc == 0) {
+      install_fd();
+      on_tcp_connected();
+    } else if (errno == EINPROGRESS) {
+      install_fd();
+      loop.mod_fd(fd, EPOLLOUT);
+    } else {
+      on_connect_failed("connect errno");
+    }
+  }
+
+  void install_fd() {
+    int f = fd;
+    loop.add_fd(fd, EPOLLIN, [this, f](uint32_t ev) {
+      if (f != fd) return;  // stale event for an already-replaced socket
+      on_socket_event(ev);
+    });
+  }
+
+  void on_socket_event(uint32_t ev) {
+    if (phase == Phase::TcpConnecting) {
+      if (ev & (EPOLLERR | EPOLLHUP)) {
+        on_connect_failed("tcp error");
+        return;
+      }
+      if (ev & EPOLLOUT) {
+        int err = 0;
+        socklen_t len = sizeof(err);
+        getsockopt(fd, SOL_SOCKET, SO_ERROR, &err, &len);
+        if (err != 0) {
+          on_connect_failed("tcp error");
+          return;
+        }
+        loop.mod_fd(fd, EPOLLIN);
+        on_tcp_connected();
+      }
+      return;
+    }
+    if (ev & (EPOLLERR | EPOLLHUP)) {
+      on_connection_lost("socket error");
+      return;
+    }
+    if (ev & EPOLLIN) {
+      if (!read_frames()) return;
+    }
+    if (ev & EPOLLOUT) flush(true);
+  }
+
+  void on_tcp_connected() {
+    phase = Phase::Handshaking;
+    ConnectRequest req;
+    req.last_zxid_seen = have_session ? last_zxid : 0;
+    req.time_out_ms = cfg.session_timeout_ms;
+    req.session_id = have_session ? sid : 0;
+    req.passwd = have_session ? passwd : std::string(16, '\0');
+    std::string pkt;
+    begin_packet(&pkt);
+    JuteWriter w(&pkt);
+    req.serialize(w);
+    frame_packet(&pkt);
+    outbuf += pkt;
+    flush(false);
+  }
+
+  void on_connect_failed(const char* why) {
+    teardown_socket();
+    if (closed.load()) return;
+    Backoff& bo = have_session ? reconnect_backoff : connect_backoff;
+    if (bo.exhausted()) {
+      log.error("zookeeper: connect attempts exhausted", {{"reason", Json(why)}});
+      phase = Phase::Stopped;
+      state.store(SessionState::Closed);
+      emit({SessionEvent::Type::Closed, 0, 0, 0});
+      return;
+    }
+    int64_t delay = bo.next_delay();
+    int64_t attempt = bo.attempt;  // 1-based after next_delay()
+    // per-attempt log level escalates info→warn→error (reference lib/zk.js:104-119)
+    std::vector<JsonMember> fields{{"attempt", Json(attempt)}, {"delay", Json(delay)}, {"reason", Json(why)},
+                                   {"server", Json(current_server)}};
+    if (attempt <= 1)
+      log.info("zookeeper: connection attempted (failed)", fields);
+    else if (attempt < 5)
+      log.warn("zookeeper: connection attempted (failed)", fields);
+    else
+      log.error("zookeeper: connection attempted (failed)", fields);
+    emit({SessionEvent::Type::ConnectAttempt, 0, attempt, delay});
+    phase = Phase::Idle;
+    retry_timer = loop.schedule(delay, [this] {
+      retry_timer = 0;
+      start_connect();
+    });
+  }
+
+  void on_connection_lost(const char* why) {
+    log.warn("zookeeper: disconnected", {{"reason", Json(why)}});
+    teardown_socket();
+    fail_all_pending(kZConnectionLoss);
+    if (closed.load()) return;
+    state.store(SessionState::Connecting);
+    emit({SessionEvent::Type::Disconnected, sid, 0, 0});
+    reconnect_backoff.reset();
+    phase = Phase::Idle;
+    start_connect();
+  }
+
+  void teardown_socket() {
+    if (conn_timer) {
+      loop.cancel(conn_timer);
+      conn_timer = 0;
+    }
+    if (ping_timer) {
+      loop.cancel(ping_timer);
+      ping_timer = 0;
+    }
+    if (fd >= 0) {
+      loop.del_fd(fd);
+      ::close(fd);
+      fd = -1;
+    }
+    inbuf.clear();
+    inpos = 0;
+    outbuf.clear();
+    flush_scheduled = false;
+  }
+
+  void fail_all_pending(int rc) {
+    std::deque<Pending> q;
+    q.swap(pending);
+    for (auto& p : q)
+      if (p.done) p.done(rc, nullptr);
+  }
+
+  // ---------------- frame processing ----------------
+
+  bool read_frames() {
+    char buf[65536];
+    while (true) {
+      ssize_t n = read(fd, buf, sizeof(buf));
+      if (n > 0) {
+        inbuf.append(buf, static_cast<size_t>(n));
+      } else if (n == 0) {
+        on_connection_lost("eof");
+        return false;
+      } else {
+        if (errno == EAGAIN || errno == EWOULDBLOCK) break;
+        if (errno == EINTR) continue;
+        on_connection_lost("read error");
+        return false;
+      }
+    }
+    last_recv = now_ms();
+    while (true) {
+      size_t avail = inbuf.size() - inpos;
+      if (avail < 4) break;
+      const unsigned char* p = reinterpret_cast<const unsigned char*>(inbuf.data() + inpos);
+      uint32_t len = (static_cast<uint32_t>(p[0]) << 24) | (static_cast<uint32_t>(p[1]) << 16) |
+                     (static_cast<uint32_t>(p[2]) << 8) | static_cast<uint32_t>(p[3]);
+      if (len > 64 * 1024 * 1024) {
+        on_connection_lost("oversized frame");
+        return false;
+      }
+      if (avail < 4 + len) break;
+      std::string body(inbuf.data() + inpos + 4, len);
+      inpos += 4 + len;
+      if (!handle_frame(body)) return false;  // handler may reconnect/teardown
+    }
+    if (inpos > 0 && fd >= 0) {
+      inbuf.erase(0, inpos);
+      inpos = 0;
+    }
+    return true;
+  }
+
+  bool handle_frame(const std::string& body) {
+    try {
+      JuteReader r(body);
+      if (phase == Phase::Handshaking) {
+        ConnectResponse resp;
+        resp.deserialize(r);
+        if (conn_timer) {
+          loop.cancel(conn_timer);
+          conn_timer = 0;
+        }
+        if (resp.session_id == 0 || resp.time_out_ms <= 0) {
+          if (have_session) {
+            log.warn("zookeeper: session expired by server", {{"session", Json(sid)}});
+            teardown_socket();
+            fail_all_pending(kZSessionExpired);
+            phase = Phase::Stopped;
+            state.store(SessionState::Expired);
+            emit({SessionEvent::Type::Expired, sid, 0, 0});
+          } else {
+            on_connect_failed("handshake rejected");
+          }
+          return false;
+        }
+        bool reconnected = have_session;
+        have_session = true;
+        sid = resp.session_id;
+        passwd = resp.passwd;
+        negotiated_timeout = resp.time_out_ms;
+        session_id_pub.store(sid);
+        negotiated_pub.store(negotiated_timeout);
+        phase = Phase::Ready;
+        connect_backoff.reset();
+        reconnect_backoff.reset();
+        last_recv = now_ms();
+        schedule_ping();
+        state.store(SessionState::Connected);
+        log.info(reconnected ? "zookeeper: reconnected" : "ZK: connected",
+                 {{"server", Json(current_server)},
+                  {"session", Json(sid)},
+                  {"timeout_ms", Json(static_cast<int64_t>(negotiated_timeout))}});
+        emit({SessionEvent::Type::Connected, sid, 0, 0});
+        return true;
+      }
+
+      ReplyHeader hdr;
+      hdr.deserialize(r);
+      if (hdr.zxid > 0) last_zxid = hdr.zxid;
+      if (hdr.xid == kXidWatcherEvent) {
+        WatcherEvent ev;
+        ev.deserialize(r);
+        log.debug("watch event", {{"path", Json(ev.path)}, {"type", Json(static_cast<int64_t>(ev.type))}});
+        if (watch_cb) watch_cb(ev);
+        return true;
+      }
+      if (pending.empty()) {
+        on_connection_lost("unexpected reply");
+        return false;
+      }
+      Pending p = std::move(pending.front());
+      pending.pop_front();
+      if (p.xid != hdr.xid) {
+        log.error("zookeeper: xid mismatch", {{"expected", Json(static_cast<int64_t>(p.xid))},
+                                              {"got", Json(static_cast<int64_t>(hdr.xid))}});
+        if (p.done) p.done(kZConnectionLoss, nullptr);
+        on_connection_lost("xid mismatch");
+        return false;
+      }
+      if (p.done) {
+        if (hdr.err == kZOk)
+          p.done(kZOk, &r);
+        else
+          p.done(hdr.err, nullptr);
+      }
+      return true;
+    } catch (const std::exception& e) {
+      log.error("zookeeper: malformed frame", {{"err", Json(e.what())}});
+      on_connection_lost("malformed frame");
+      return false;
+    }
+  }
+
+  // ---------------- ping ----------------
+
+  void schedule_ping() {
+    int64_t interval = std::max<int64_t>(negotiated_timeout / 3, 100);
+    ping_timer = loop.schedule(interval, [this] {
+      ping_timer = 0;
+      if (phase != Phase::Ready) return;
+      // server silent for 2/3 of the session timeout ⇒ presume dead, move on
+      if (now_ms() - last_recv > negotiated_timeout * 2 / 3) {
+        on_connection_lost("ping timeout");
+        return;
+      }
+      submit(kXidPing, kOpPing, [](JuteWriter&) {}, nullptr);
+      schedule_ping();
+    });
+  }
+
+  // ---------------- request submission (loop thread) ----------------
+
+  template <typename SerFn>
+  void submit(int32_t xid, int32_t op, SerFn serialize_body, std::function<void(int, JuteReader*)> done) {
+    if (phase != Phase::Ready) {
+      if (done) done(phase == Phase::Stopped ? kZSessionExpired : kZConnectionLoss, nullptr);
+      return;
+    }
+    std::string pkt;
+    begin_packet(&pkt);
+    JuteWriter w(&pkt);
+    RequestHeader hdr;
+    hdr.xid = xid;
+    hdr.type = op;
+    hdr.serialize(w);
+    serialize_body(w);
+    frame_packet(&pkt);
+    outbuf += pkt;
+    pending.push_back(Pending{xid, op, std::move(done)});
+    schedule_flush();
+  }
+
+  template <typename SerFn>
+  void submit_op(int32_t op, SerFn serialize_body, std::function<void(int, JuteReader*)> done) {
+    submit(next_xid++, op, serialize_body, std::move(done));
+  }
+
+  // Batch all writes queued in this loop-drain cycle into one flush: every
+  // async op posted from other threads lands in the same drain, so a
+  // register() of 1k nodes goes out as a handful of large TCP writes instead
+  // of 1k small ones.
+  void schedule_flush() {
+    if (flush_scheduled) return;
+    flush_scheduled = true;
+    loop.schedule(0, [this] {
+      flush_scheduled = false;
+      flush(false);
+    });
+  }
+
+  void flush(bool from_epollout) {
+    if (fd < 0) return;
+    size_t off = 0;
+    while (off < outbuf.size()) {
+      ssize_t n = write(fd, outbuf.data() + off, outbuf.size() - off);
+      if (n > 0) {
+        off += static_cast<size_t>(n);
+      } else if (n < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) {
+        break;
+      } else if (n < 0 && errno == EINTR) {
+        continue;
+      } else {
+        on_connection_lost("write error");
+        return;
+      }
+    }
+    outbuf.erase(0, off);
+    if (!outbuf.empty()) {
+      loop.mod_fd(fd, EPOLLIN | EPOLLOUT);
+    } else if (from_epollout) {
+      loop.mod_fd(fd, EPOLLIN);
+    }
+  }
+};
+
+// ---------------- public API ----------------
+
+ZkClient::ZkClient(ZkClientConfig cfg, Logger log) : impl_(std::make_unique<Impl>(std::move(cfg), log)) {
+  if (impl_->cfg.servers.empty()) throw std::runtime_error("ZkClient: options.servers empty");
+}
+
+ZkClient::~ZkClient() {
+  try {
+    close();
+  } catch (...) {
+  }
+}
+
+void ZkClient::start() {
+  if (impl_->started.exchange(true)) return;
+  impl_->thread = std::thread([this] {
+    impl_->loop.post([this] { impl_->start_connect(); });
+    impl_->loop.run();
+  });
+}
+
+bool ZkClient::wait_connected(int64_t timeout_ms) {
+  std::unique_lock<std::mutex> g(impl_->ev_mu);
+  auto pred = [this] { return impl_->settled; };
+  if (timeout_ms < 0) {
+    impl_->ev_cv.wait(g, pred);
+  } else {
+    if (!impl_->ev_cv.wait_for(g, std::chrono::milliseconds(timeout_ms), pred)) return false;
+  }
+  return impl_->state.load() == SessionState::Connected;
+}
+
+void ZkClient::abort_connect() {
+  impl_->closed.store(true);
+  impl_->loop.post([this] {
+    if (impl_->retry_timer) {
+      impl_->loop.cancel(impl_->retry_timer);
+      impl_->retry_timer = 0;
+    }
+    impl_->teardown_socket();
+    impl_->phase = Impl::Phase::Stopped;
+    impl_->state.store(SessionState::Closed);
+    impl_->emit({SessionEvent::Type::Closed, 0, 0, 0});
+  });
+}
+
+void ZkClient::close() {
+  if (!impl_->started.load()) return;
+  if (!impl_->closed.exchange(true)) {
+    std::promise<void> done;
+    impl_->loop.post([this, &done] {
+      if (impl_->phase == Impl::Phase::Ready) {
+        // best-effort closeSession then synchronous drain
+        impl_->submit(impl_->next_xid++, kOpCloseSession, [](JuteWriter&) {}, nullptr);
+        impl_->flush(false);
+      }
+      impl_->fail_all_pending(kZConnectionLoss);
+      if (impl_->retry_timer) impl_->loop.cancel(impl_->retry_timer);
+      impl_->teardown_socket();
+      impl_->phase = Impl::Phase::Stopped;
+      impl_->state.store(SessionState::Closed);
+      impl_->emit({SessionEvent::Type::Closed, impl_->sid, 0, 0});
+      done.set_value();
+    });
+    done.get_future().wait();
+  }
+  impl_->loop.stop();
+  if (impl_->thread.joinable()) impl_->thread.join();
+  impl_->started.store(false);
+}
+
+SessionState ZkClient::state() const { return impl_->state.load(); }
+int64_t ZkClient::session_id() const { return impl_->session_id_pub.load(); }
+int64_t ZkClient::session_timeout_ms() const { return impl_->negotiated_pub.load(); }
+
+std::string ZkClient::to_string() const {
+  std::string s = "zk://";
+  for (size_t i = 0; i < impl_->cfg.servers.size(); i++) {
+    if (i) s += ',';
+    s += impl_->cfg.servers[i].host + ":" + std::to_string(impl_->cfg.servers[i].port);
+  }
+  char buf[32];
+  snprintf(buf, sizeof(buf), " (session 0x%llx)", static_cast<unsigned long long>(session_id()));
+  s += buf;
+  return s;
+}
+
+void ZkClient::set_event_callback(EventCallback cb) { impl_->ev_cb = std::move(cb); }
+void ZkClient::set_watch_callback(WatchCallback cb) { impl_->watch_cb = std::move(cb); }
+
+std::vector<SessionEvent> ZkClient::poll_events() {
+  std::lock_guard<std::mutex> g(impl_->ev_mu);
+  std::vector<SessionEvent> out;
+  out.swap(impl_->ev_queue);
+  return out;
+}
+
+// ---- async ops ----
+
+void ZkClient::acreate(const std::string& path, const std::string& data, int32_t flags, StringCallback cb) {
+  impl_->loop.post([this, path, data, flags, cb = std::move(cb)] {
+    CreateRequest req;
+    req.path = path;
+    req.data = data;
+    req.flags = flags;
+    impl_->submit_op(kOpCreate, [req](JuteWriter& w) { req.serialize(w); },
+                     [cb](int rc, JuteReader* r) {
+                       if (!cb) return;
+                       std::string created;
+                       if (rc == kZOk && r) {
+                         CreateResponse resp;
+                         resp.deserialize(*r);
+                         created = resp.path;
+                       }
+                       cb(rc, created);
+                     });
+  });
+}
+
+void ZkClient::adelete(const std::string& path, int32_t version, VoidCallback cb) {
+  impl_->loop.post([this, path, version, cb = std::move(cb)] {
+    DeleteRequest req;
+    req.path = path;
+    req.version = version;
+    impl_->submit_op(kOpDelete, [req](JuteWriter& w) { req.serialize(w); },
+                     [cb](int rc, JuteReader*) {
+                       if (cb) cb(rc);
+                     });
+  });
+}
+
+void ZkClient::aexists(const std::string& path, bool watch, StatCallback cb) {
+  impl_->loop.post([this, path, watch, cb = std::move(cb)] {
+    ExistsRequest req;
+    req.path = path;
+    req.watch = watch;
+    impl_->submit_op(kOpExists, [req](JuteWriter& w) { req.serialize(w); },
+                     [cb](int rc, JuteReader* r) {
+                       if (!cb) return;
+                       Stat st;
+                       if (rc == kZOk && r) {
+                         ExistsResponse resp;
+                         resp.deserialize(*r);
+                         st = resp.stat;
+                       }
+                       cb(rc, st);
+                     });
+  });
+}
+
+void ZkClient::aget(const std::string& path, bool watch, DataCallback cb) {
+  impl_->loop.post([this, path, watch, cb = std::move(cb)] {
+    GetDataRequest req;
+    req.path = path;
+    req.watch = watch;
+    impl_->submit_op(kOpGetData, [req](JuteWriter& w) { req.serialize(w); },
+                     [cb](int rc, JuteReader* r) {
+                       if (!cb) return;
+                       std::string data;
+                       Stat st;
+                       if (rc == kZOk && r) {
+                         GetDataResponse resp;
+                         resp.deserialize(*r);
+                         data = resp.data;
+                         st = resp.stat;
+                       }
+                       cb(rc, data, st);
+                     });
+  });
+}
+
+void ZkClient::aset(const std::string& path, const std::string& data, int32_t version, StatCallback cb) {
+  impl_->loop.post([this, path, data, version, cb = std::move(cb)] {
+    SetDataRequest req;
+    req.path = path;
+    req.data = data;
+    req.version = version;
+    impl_->submit_op(kOpSetData, [req](JuteWriter& w) { req.serialize(w); },
+                     [cb](int rc, JuteReader* r) {
+                       if (!cb) return;
+                       Stat st;
+                       if (rc == kZOk && r) {
+                         SetDataResponse resp;
+                         resp.deserialize(*r);
+                         st = resp.stat;
+                       }
+                       cb(rc, st);
+                     });
+  });
+}
+
+void ZkClient::achildren(const std::string& path, bool watch, ChildrenCallback cb) {
+  impl_->loop.post([this, path, watch, cb = std::move(cb)] {
+    GetChildrenRequest req;
+    req.path = path;
+    req.watch = watch;
+    impl_->submit_op(kOpGetChildren, [req](JuteWriter& w) { req.serialize(w); },
+                     [cb](int rc, JuteReader* r) {
+                       if (!cb) return;
+                       std::vector<std::string> children;
+                       if (rc == kZOk && r) {
+                         GetChildrenResponse resp;
+                         resp.deserialize(*r);
+                         children = std::move(resp.children);
+                       }
+                       cb(rc, children);
+                     });
+  });
+}
+
+// ---- sync wrappers ----
+
+int ZkClient::create(const std::string& path, const std::string& data, int32_t flags, std::string* created_path) {
+  std::promise<std::pair<int, std::string>> p;
+  acreate(path, data, flags, [&p](int rc, const std::string& cp) { p.set_value({rc, cp}); });
+  auto [rc, cp] = p.get_future().get();
+  if (created_path) *created_path = cp;
+  return rc;
+}
+
+int ZkClient::del(const std::string& path, int32_t version) {
+  std::promise<int> p;
+  adelete(path, version, [&p](int rc) { p.set_value(rc); });
+  return p.get_future().get();
+}
+
+int ZkClient::exists(const std::string& path, Stat* stat) {
+  std::promise<std::pair<int, Stat>> p;
+  aexists(path, false, [&p](int rc, const Stat& st) { p.set_value({rc, st}); });
+  auto [rc, st] = p.get_future().get();
+  if (stat) *stat = st;
+  return rc;
+}
+
+int ZkClient::get(const std::string& path, std::string* data, Stat* stat) {
+  std::promise<int> p;
+  aget(path, false, [&](int rc, const std::string& d, const Stat& st) {
+    if (data) *data = d;
+    if (stat) *stat = st;
+    p.set_value(rc);
+  });
+  return p.get_future().get();
+}
+
+int ZkClient::set(const std::string& path, const std::string& data, int32_t version, Stat* stat) {
+  std::promise<int> p;
+  aset(path, data, version, [&](int rc, const Stat& st) {
+    if (stat) *stat = st;
+    p.set_value(rc);
+  });
+  return p.get_future().get();
+}
+
+int ZkClient::get_children(const std::string& path, std::vector<std::string>* children) {
+  std::promise<int> p;
+  achildren(path, false, [&](int rc, const std::vector<std::string>& ch) {
+    if (children) *children = ch;
+    p.set_value(rc);
+  });
+  return p.get_future().get();
+}
+
+// ---- zkplus-surface verbs ----
+
+int ZkClient::put(const std::string& path, const std::string& data) {
+  // create-or-overwrite persistent node (zkplus `put`, used for the service
+  // record at lib/register.js:62)
+  int rc = set(path, data, -1);
+  if (rc == kZNoNode) {
+    rc = create(path, data, 0);
+    if (rc == kZNodeExists) rc = set(path, data, -1);  // lost the race; overwrite
+  }
+  return rc;
+}
+
+int ZkClient::mkdirp(const std::string& path) {
+  if (path.empty() || path[0] != '/') return kZMarshallingError;
+  if (path == "/") return kZOk;
+  // Pipeline the whole prefix chain in one flush: in-order processing
+  // guarantees each parent exists (or already existed) by the time its child
+  // create is handled.
+  std::vector<std::string> prefixes;
+  size_t pos = 0;
+  while ((pos = path.find('/', pos + 1)) != std::string::npos) prefixes.push_back(path.substr(0, pos));
+  prefixes.push_back(path);
+  std::vector<std::string> datas(prefixes.size());
+  std::vector<int> rcs = create_many(prefixes, datas, 0);
+  for (int rc : rcs)
+    if (rc != kZOk && rc != kZNodeExists) return rc;
+  return kZOk;
+}
+
+int ZkClient::unlink(const std::string& path) { return del(path, -1); }
+
+// ---- pipelined batches ----
+
+namespace {
+struct BatchState {
+  std::mutex mu;
+  std::condition_variable cv;
+  size_t done = 0;
+  size_t total = 0;
+};
+}  // namespace
+
+std::vector<int> ZkClient::create_many(const std::vector<std::string>& paths, const std::vector<std::string>& datas,
+                                       int32_t flags) {
+  size_t n = paths.size();
+  std::vector<int> rcs(n, kZConnectionLoss);
+  if (n == 0) return rcs;
+  auto st = std::make_shared<BatchState>();
+  st->total = n;
+  impl_->loop.post([this, &paths, &datas, &rcs, flags, st, n] {
+    for (size_t i = 0; i < n; i++) {
+      CreateRequest req;
+      req.path = paths[i];
+      req.data = datas[i];
+      req.flags = flags;
+      impl_->submit_op(kOpCreate, [req](JuteWriter& w) { req.serialize(w); },
+                       [st, &rcs, i](int rc, JuteReader*) {
+                         std::lock_guard<std::mutex> g(st->mu);
+                         rcs[i] = rc;
+                         if (++st->done == st->total) st->cv.notify_all();
+                       });
+    }
+  });
+  std::unique_lock<std::mutex> g(st->mu);
+  st->cv.wait(g, [&] { return st->done == st->total; });
+  return rcs;
+}
+
+std::vector<int> ZkClient::delete_many(const std::vector<std::string>& paths) {
+  size_t n = paths.size();
+  std::vector<int> rcs(n, kZConnectionLoss);
+  if (n == 0) return rcs;
+  auto st = std::make_shared<BatchState>();
+  st->total = n;
+  impl_->loop.post([this, &paths, &rcs, st, n] {
+    for (size_t i = 0; i < n; i++) {
+      DeleteRequest req;
+      req.path = paths[i];
+      req.version = -1;
+      impl_->submit_op(kOpDelete, [req](JuteWriter& w) { req.serialize(w); },
+                       [st, &rcs, i](int rc, JuteReader*) {
+                         std::lock_guard<std::mutex> g(st->mu);
+                         rcs[i] = rc;
+                         if (++st->done == st->total) st->cv.notify_all();
+                       });
+    }
+  });
+  std::unique_lock<std::mutex> g(st->mu);
+  st->cv.wait(g, [&] { return st->done == st->total; });
+  return rcs;
+}
+
+std::vector<int> ZkClient::exists_many(const std::vector<std::string>& paths, std::vector<Stat>* stats) {
+  size_t n = paths.size();
+  std::vector<int> rcs(n, kZConnectionLoss);
+  if (stats) stats->assign(n, Stat{});
+  if (n == 0) return rcs;
+  auto st = std::make_shared<BatchState>();
+  st->total = n;
+  impl_->loop.post([this, &paths, &rcs, stats, st, n] {
+    for (size_t i = 0; i < n; i++) {
+      ExistsRequest req;
+      req.path = paths[i];
+      req.watch = false;
+      impl_->submit_op(kOpExists, [req](JuteWriter& w) { req.serialize(w); },
+                       [st, &rcs, stats, i](int rc, JuteReader* r) {
+                         if (rc == kZOk && r && stats) {
+                           ExistsResponse resp;
+                           resp.deserialize(*r);
+                           (*stats)[i] = resp.stat;
+                         }
+                         std::lock_guard<std::mutex> g(st->mu);
+                         rcs[i] = rc;
+                         if (++st->done == st->total) st->cv.notify_all();
+                       });
+    }
+  });
+  std::unique_lock<std::mutex> g(st->mu);
+  st->cv.wait(g, [&] { return st->done == st->total; });
+  return rcs;
+}
+
+int ZkClient::heartbeat(const std::vector<std::string>& nodes, const RetryPolicy& retry, int64_t* rtt_us) {
+  Backoff bo;
+  bo.initial_ms = retry.initial_delay_ms;
+  bo.max_ms = retry.max_delay_ms;
+  bo.max_attempts = retry.max_attempts;
+  while (true) {
+    int64_t t0 = now_us();
+    std::vector<int> rcs = exists_many(nodes, nullptr);
+    int rc = kZOk;
+    for (int r : rcs) {
+      if (r != kZOk) {
+        rc = r;
+        break;
+      }
+    }
+    if (rc == kZOk) {
+      if (rtt_us) *rtt_us = now_us() - t0;
+      return kZOk;
+    }
+    if (bo.exhausted()) return rc;
+    int64_t delay = bo.next_delay();
+    impl_->log.debug("heartbeat: retrying", {{"rc", Json(error_name(rc))}, {"delay_ms", Json(delay)}});
+    std::this_thread::sleep_for(std::chrono::milliseconds(delay));
+  }
+}
+
+}  // namespace zk
+}  // namespace registrar
