@@ -229,22 +229,29 @@ struct Ensemble::Impl {
       return;
     }
     if (ev & EPOLLIN) {
+      bool eof = false;
       char buf[65536];
       while (true) {
         ssize_t n = read(c->fd, buf, sizeof(buf));
         if (n > 0) {
           c->inbuf.append(buf, static_cast<size_t>(n));
         } else if (n == 0) {
-          close_conn_locked(c);
-          return;
+          eof = true;  // process buffered frames (e.g. closeSession) first
+          break;
         } else {
           if (errno == EAGAIN || errno == EWOULDBLOCK) break;
           if (errno == EINTR) continue;
-          close_conn_locked(c);
-          return;
+          eof = true;
+          break;
         }
       }
+      uint64_t cid = c->id;
       if (!process_frames_locked(c)) return;  // conn closed
+      if (eof) {
+        auto it2 = conns.find(cid);
+        if (it2 != conns.end()) close_conn_locked(it2->second.get());
+        return;
+      }
     }
     if (ev & EPOLLOUT) flush_out_locked(c);
   }

@@ -536,6 +536,7 @@ void ZkClient::abort_connect() {
       impl_->retry_timer = 0;
     }
     impl_->teardown_socket();
+    impl_->fail_all_pending(zk::kZConnectionLoss);
     impl_->phase = Impl::Phase::Stopped;
     impl_->state.store(SessionState::Closed);
     impl_->emit({SessionEvent::Type::Closed, 0, 0, 0});
@@ -900,6 +901,10 @@ int ZkClient::heartbeat(const std::vector<std::string>& nodes, const RetryPolicy
       if (rtt_us) *rtt_us = now_us() - t0;
       return kZOk;
     }
+    // terminal states: retrying cannot help, and sleeping here would stall
+    // the owner's expiry handling (the orchestrator must re-register NOW)
+    SessionState st = state();
+    if (rc == kZSessionExpired || st == SessionState::Expired || st == SessionState::Closed) return rc;
     if (bo.exhausted()) return rc;
     int64_t delay = bo.next_delay();
     impl_->log.debug("heartbeat: retrying", {{"rc", Json(error_name(rc))}, {"delay_ms", Json(delay)}});
