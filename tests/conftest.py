@@ -67,11 +67,9 @@ def free_port():
 
 @pytest.fixture(scope="session")
 def daemon_bin():
-    """Path to the registrard binary, building it if needed."""
-    path = os.path.join(REPO_ROOT, "bin", "registrard")
-    if not os.path.exists(path):
-        subprocess.run(["make", "daemon"], cwd=REPO_ROOT, check=True, capture_output=True)
-    return path
+    """Path to the registrard binary; make keeps it current vs csrc/."""
+    subprocess.run(["make", "daemon"], cwd=REPO_ROOT, check=True, capture_output=True)
+    return os.path.join(REPO_ROOT, "bin", "registrard")
 
 
 def orch_config(ens, registration, **extra):

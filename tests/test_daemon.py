@@ -1,0 +1,127 @@
+"""registrard daemon-level integration: spawn the real binary against an
+in-process ensemble and observe its bunyan log stream and ZK effects.
+
+The reference has no daemon-level tests at all; its SMF manifest contract
+(exit on expiry → restarter relaunches) is verified here
+with --exit-on-expiry."""
+import json
+import os
+import signal
+import subprocess
+import time
+
+import pytest
+
+import registrar_amd as ra
+from conftest import REPO_ROOT, orch_config, wait_for
+
+
+def spawn_daemon(daemon_bin, cfg, *args):
+    cfg_path = os.path.join("/tmp", "registrard-test-%d.json" % os.getpid())
+    with open(cfg_path, "w") as f:
+        json.dump(cfg, f)
+    proc = subprocess.Popen(
+        [daemon_bin, "-f", cfg_path, *args],
+        stdout=subprocess.PIPE,
+        stderr=subprocess.STDOUT,
+        text=True,
+    )
+    return proc, cfg_path
+
+
+def read_logs(proc):
+    out, _ = proc.communicate(timeout=10)
+    recs = []
+    for line in out.splitlines():
+        try:
+            recs.append(json.loads(line))
+        except ValueError:
+            pass
+    return recs
+
+
+def test_daemon_registers_and_heartbeats(ensemble, daemon_bin):
+    cfg = orch_config(
+        ensemble,
+        {"domain": "d.test", "type": "host", "hostname": "dh", "adminIp": "127.0.0.1", "settleMs": 0},
+        heartbeatInterval=100,
+    )
+    proc, cfg_path = spawn_daemon(daemon_bin, cfg)
+    try:
+        assert wait_for(lambda: ensemble.get("/test/d/dh")["exists"], timeout=10)
+        payload = json.loads(ensemble.get("/test/d/dh")["data"])
+        assert payload == {"type": "host", "address": "127.0.0.1", "host": {"address": "127.0.0.1"}}
+        time.sleep(0.5)
+    finally:
+        proc.send_signal(signal.SIGTERM)
+        recs = read_logs(proc)
+        os.unlink(cfg_path)
+    assert proc.returncode == 0
+    msgs = [r["msg"] for r in recs]
+    assert "registrar: registered" in msgs
+    assert any(r.get("name") == "registrar" and "hostname" in r and "pid" in r and "v" in r for r in recs), \
+        "bunyan record shape missing"
+
+
+def test_daemon_exit_on_expiry(ensemble, daemon_bin):
+    cfg = orch_config(
+        ensemble,
+        {"domain": "e.test", "type": "host", "hostname": "eh", "settleMs": 0},
+        heartbeatInterval=100,
+    )
+    proc, cfg_path = spawn_daemon(daemon_bin, cfg, "--exit-on-expiry")
+    try:
+        assert wait_for(lambda: ensemble.get("/test/e/eh")["exists"], timeout=10)
+        sid = ensemble.get("/test/e/eh")["stat"]["ephemeralOwner"]
+        ensemble.expire_session(sid)
+        assert wait_for(lambda: proc.poll() is not None, timeout=15)
+    finally:
+        if proc.poll() is None:
+            proc.kill()
+        recs = read_logs(proc)
+        os.unlink(cfg_path)
+    # exit(1) on expiry, with the reference's fatal message contract
+    # (main.js:141-144)
+    assert proc.returncode == 1
+    assert any("session_expired" in r["msg"] for r in recs)
+
+
+def test_daemon_reregisters_in_process_on_expiry(ensemble, daemon_bin):
+    cfg = orch_config(
+        ensemble,
+        {"domain": "r.test", "type": "host", "hostname": "rh", "settleMs": 0},
+        heartbeatInterval=100,
+    )
+    proc, cfg_path = spawn_daemon(daemon_bin, cfg)
+    try:
+        assert wait_for(lambda: ensemble.get("/test/r/rh")["exists"], timeout=10)
+        sid1 = ensemble.get("/test/r/rh")["stat"]["ephemeralOwner"]
+        ensemble.expire_session(sid1)
+        # default policy: daemon stays up, re-registers under a new session
+        assert wait_for(
+            lambda: ensemble.get("/test/r/rh")["exists"]
+            and ensemble.get("/test/r/rh")["stat"]["ephemeralOwner"] not in (0, sid1),
+            timeout=15,
+        )
+        assert proc.poll() is None
+    finally:
+        proc.send_signal(signal.SIGTERM)
+        read_logs(proc)
+        os.unlink(cfg_path)
+    assert proc.returncode == 0
+
+
+def test_daemon_bad_config(daemon_bin):
+    proc = subprocess.Popen([daemon_bin, "-f", "/nonexistent.json"], stdout=subprocess.PIPE, text=True)
+    out, _ = proc.communicate(timeout=10)
+    assert proc.returncode == 1
+    assert "unable to read configuration" in out
+
+
+def test_daemon_usage():
+    daemon = os.path.join(REPO_ROOT, "bin", "registrard")
+    if not os.path.exists(daemon):
+        pytest.skip("daemon not built")
+    proc = subprocess.run([daemon, "-h"], capture_output=True, text=True, timeout=10)
+    assert proc.returncode == 0
+    assert "usage:" in proc.stdout
