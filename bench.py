@@ -50,7 +50,10 @@ def main():
     ap.add_argument("--steps", type=int, default=30)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--znodes", type=int, default=ZNODES_PER_PROC)
-    ap.add_argument("--servers", type=int, default=1, help="ensemble server count")
+    ap.add_argument("--servers", type=int, default=0,
+                    help="ensemble server count (0 = auto: 1 for a single proc, "
+                         "3 — the standard ZK quorum size — for multi-proc, "
+                         "matching BASELINE configs 1 vs 2/3)")
     args = ap.parse_args()
 
     import registrar_amd as ra
@@ -80,6 +83,8 @@ def main():
             torch.cuda.synchronize()
 
     # ---- ensemble: hosted by rank 0, shared over localhost TCP ----
+    if args.servers <= 0:
+        args.servers = 1 if world == 1 else 3
     ensemble = None
     if rank == 0:
         ensemble = ra.Ensemble(servers=args.servers, tick_ms=100, max_session_timeout_ms=60000)

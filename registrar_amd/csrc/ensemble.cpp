@@ -1,4 +1,12 @@
 // ensemble.cpp — synthetic in-process ZooKeeper ensemble (see ensemble.hpp).
+//
+// Threading model (the scaling core): one epoll loop THREAD PER SERVER, so N
+// registrar processes spread across the ensemble's servers get parallel
+// socket I/O and frame processing. The shared data tree / session table /
+// watch maps live under one state mutex (`mu`) whose critical sections are
+// memory-only (no syscalls); per-connection output buffers have their own
+// mutex so reply writes and cross-thread watch deliveries never serialize on
+// the global lock. Lock order: mu → conn->out_mu (never the reverse).
 #include "ensemble.hpp"
 
 #include <arpa/inet.h>
@@ -9,6 +17,7 @@
 #include <unistd.h>
 
 #include <algorithm>
+#include <array>
 #include <cstdio>
 #include <future>
 
@@ -72,13 +81,20 @@ struct Ensemble::Impl {
     uint64_t id = 0;
     int fd = -1;
     size_t server_idx = 0;
+    // owner-loop-thread only:
     std::string inbuf;
     size_t inpos = 0;
-    std::string outbuf;
     bool handshaken = false;
-    bool closing = false;  // close once outbuf drains
     int64_t session_id = 0;
+    // shared output state:
+    std::mutex out_mu;
+    std::string outbuf;
+    bool epollout_armed = false;
+    bool closing = false;  // close once outbuf drains
+    std::atomic<bool> dead{false};
+    bool flush_scheduled = false;  // owner loop thread only
   };
+  using ConnPtr = std::shared_ptr<Conn>;
 
   struct Server {
     int listen_fd = -1;
@@ -86,17 +102,23 @@ struct Ensemble::Impl {
     bool up = false;
   };
 
+  struct OpCounters {
+    std::atomic<uint64_t> connect{0}, create{0}, del{0}, exists{0}, get_data{0}, set_data{0},
+        get_children{0}, ping{0}, close_session{0}, unknown{0};
+  };
+
   EnsembleConfig cfg;
   Logger log;
-  EventLoop loop;
-  std::thread thread;
+  std::vector<std::unique_ptr<EventLoop>> loops;  // one per server slot
+  std::vector<std::thread> threads;
   std::atomic<bool> started{false};
+  std::atomic<int> latency_ms{0};
+  OpCounters ops;
 
-  // All mutable ensemble state below is guarded by mu (request handlers run
-  // on the loop thread but introspection comes from arbitrary threads).
+  // ---- state under mu ----
   mutable std::mutex mu;
   std::vector<Server> servers;
-  std::unordered_map<uint64_t, std::unique_ptr<Conn>> conns;
+  std::unordered_map<uint64_t, ConnPtr> conns;
   uint64_t next_conn_id = 1;
   std::unordered_map<std::string, ZNode> nodes;
   std::unordered_map<int64_t, Session> sessions;
@@ -108,8 +130,6 @@ struct Ensemble::Impl {
   int64_t next_session = 0x100000;
   size_t leader_idx = 0;
   int64_t election_until = 0;  // monotonic ms; connects refused until then
-  std::atomic<int> latency_ms{0};
-  std::map<std::string, uint64_t> op_counters;
 
   explicit Impl(EnsembleConfig c) : cfg(std::move(c)), log(Logger("zk-ensemble").child("ensemble")) {
     log.set_level(cfg.log_level);
@@ -121,44 +141,55 @@ struct Ensemble::Impl {
 
   void start() {
     if (started.exchange(true)) return;
+    size_t n = cfg.ports.size();
+    loops.clear();
+    for (size_t i = 0; i < n; i++) loops.push_back(std::make_unique<EventLoop>());
     {
       std::lock_guard<std::mutex> g(mu);
-      servers.resize(cfg.ports.size());
-      for (size_t i = 0; i < cfg.ports.size(); i++) open_listener(i, cfg.ports[i]);
+      servers.resize(n);
+      for (size_t i = 0; i < n; i++) open_listener(i, cfg.ports[i]);
     }
-    thread = std::thread([this] {
-      schedule_sweep();
-      loop.run();
-    });
+    for (size_t i = 0; i < n; i++) {
+      threads.emplace_back([this, i] {
+        if (i == 0) schedule_sweep();
+        loops[i]->run();
+      });
+    }
   }
 
   void stop() {
     if (!started.load()) return;
-    std::promise<void> done;
-    loop.post([this, &done] {
-      std::lock_guard<std::mutex> g(mu);
-      for (auto& kv : conns) {
-        loop.del_fd(kv.second->fd);
-        close(kv.second->fd);
-      }
-      conns.clear();
-      for (auto& s : servers) {
-        if (s.listen_fd >= 0) {
-          loop.del_fd(s.listen_fd);
-          close(s.listen_fd);
-          s.listen_fd = -1;
-          s.up = false;
+    // close everything from each owner loop, then stop the loops
+    std::vector<std::promise<void>> done(loops.size());
+    for (size_t i = 0; i < loops.size(); i++) {
+      loops[i]->post([this, i, &done] {
+        std::vector<ConnPtr> victims;
+        {
+          std::lock_guard<std::mutex> g(mu);
+          for (auto& kv : conns)
+            if (kv.second->server_idx == i) victims.push_back(kv.second);
         }
-      }
-      done.set_value();
-    });
-    done.get_future().wait();
-    loop.stop();
-    if (thread.joinable()) thread.join();
+        for (auto& c : victims) close_conn(c.get());
+        {
+          std::lock_guard<std::mutex> g(mu);
+          if (i < servers.size() && servers[i].listen_fd >= 0) {
+            loops[i]->del_fd(servers[i].listen_fd);
+            ::close(servers[i].listen_fd);
+            servers[i].listen_fd = -1;
+            servers[i].up = false;
+          }
+        }
+        done[i].set_value();
+      });
+    }
+    for (auto& d : done) d.get_future().wait();
+    for (auto& l : loops) l->stop();
+    for (auto& t : threads) t.join();
+    threads.clear();
     started.store(false);
   }
 
-  // mu held
+  // mu held; listener registration goes to loops[idx]
   void open_listener(size_t idx, int port) {
     int fd = socket(AF_INET, SOCK_STREAM | SOCK_CLOEXEC, 0);
     if (fd < 0) throw std::runtime_error("ensemble: socket() failed");
@@ -170,11 +201,11 @@ struct Ensemble::Impl {
     addr.sin_port = htons(static_cast<uint16_t>(port));
     inet_pton(AF_INET, cfg.bind_host.c_str(), &addr.sin_addr);
     if (bind(fd, reinterpret_cast<struct sockaddr*>(&addr), sizeof(addr)) < 0) {
-      close(fd);
+      ::close(fd);
       throw std::runtime_error("ensemble: bind failed on port " + std::to_string(port));
     }
     if (listen(fd, 512) < 0) {
-      close(fd);
+      ::close(fd);
       throw std::runtime_error("ensemble: listen failed");
     }
     socklen_t alen = sizeof(addr);
@@ -184,48 +215,55 @@ struct Ensemble::Impl {
     servers[idx].port = ntohs(addr.sin_port);
     servers[idx].up = true;
     size_t srv = idx;
-    // add_fd must run on the loop thread once it is running; during start()
-    // the loop thread hasn't started yet, so direct add is safe. After
-    // restart_server we always go through loop.post.
-    auto install = [this, fd, srv] { loop.add_fd(fd, EPOLLIN, [this, fd, srv](uint32_t) { on_accept(fd, srv); }); };
-    if (started.load() && !loop.on_loop_thread()) {
-      loop.post(install);
-    } else {
+    auto install = [this, fd, srv] {
+      loops[srv]->add_fd(fd, EPOLLIN, [this, fd, srv](uint32_t) { on_accept(fd, srv); });
+    };
+    if (loops[srv]->on_loop_thread()) {
       install();
+    } else if (started.load() && loops[srv]->running()) {
+      loops[srv]->post(install);
+    } else {
+      install();  // loop thread not started yet: direct registration is safe
     }
   }
 
-  // ---------------- socket handling (loop thread) ----------------
+  // ---------------- socket handling (owner loop thread) ----------------
 
   void on_accept(int listen_fd, size_t server_idx) {
     while (true) {
       int fd = accept4(listen_fd, nullptr, nullptr, SOCK_NONBLOCK | SOCK_CLOEXEC);
       if (fd < 0) break;
-      std::lock_guard<std::mutex> g(mu);
-      if (now_ms() < election_until) {
-        // mid-election: nobody serves (BASELINE config 4 storm realism)
-        close(fd);
-        continue;
+      {
+        std::lock_guard<std::mutex> g(mu);
+        if (now_ms() < election_until) {
+          // mid-election: nobody serves (BASELINE config 4 storm realism)
+          ::close(fd);
+          continue;
+        }
+        maybe_elect_leader_locked();
+        set_nodelay(fd);
+        auto conn = std::make_shared<Conn>();
+        conn->id = next_conn_id++;
+        conn->fd = fd;
+        conn->server_idx = server_idx;
+        conns[conn->id] = conn;
+        uint64_t cid = conn->id;
+        loops[server_idx]->add_fd(fd, EPOLLIN, [this, cid](uint32_t ev) { on_conn_event(cid, ev); });
       }
-      maybe_elect_leader_locked();
-      set_nodelay(fd);
-      auto conn = std::make_unique<Conn>();
-      conn->id = next_conn_id++;
-      conn->fd = fd;
-      conn->server_idx = server_idx;
-      uint64_t cid = conn->id;
-      conns[cid] = std::move(conn);
-      loop.add_fd(fd, EPOLLIN, [this, cid](uint32_t ev) { on_conn_event(cid, ev); });
     }
   }
 
-  void on_conn_event(uint64_t cid, uint32_t ev) {
-    std::unique_lock<std::mutex> g(mu);
+  ConnPtr lookup(uint64_t cid) {
+    std::lock_guard<std::mutex> g(mu);
     auto it = conns.find(cid);
-    if (it == conns.end()) return;
-    Conn* c = it->second.get();
-    if (ev & (EPOLLHUP | EPOLLERR)) {
-      close_conn_locked(c);
+    return it == conns.end() ? nullptr : it->second;
+  }
+
+  void on_conn_event(uint64_t cid, uint32_t ev) {
+    ConnPtr c = lookup(cid);
+    if (!c) return;
+    if (c->dead.load() || (ev & (EPOLLHUP | EPOLLERR))) {
+      close_conn(c.get());
       return;
     }
     if (ev & EPOLLIN) {
@@ -245,41 +283,39 @@ struct Ensemble::Impl {
           break;
         }
       }
-      uint64_t cid = c->id;
-      if (!process_frames_locked(c)) return;  // conn closed
+      if (!process_frames(c.get())) return;
+      // replies queued during this drain go out in one flush (syscall
+      // batching: 1k pipelined creates ⇒ a handful of large writes)
+      flush_out(c.get());
+      if (c->dead.load()) return;  // flush_out may close on error/closing
       if (eof) {
-        auto it2 = conns.find(cid);
-        if (it2 != conns.end()) close_conn_locked(it2->second.get());
+        close_conn(c.get());
         return;
       }
+      return;
     }
-    if (ev & EPOLLOUT) flush_out_locked(c);
+    if (ev & EPOLLOUT) flush_out(c.get());
   }
 
-  // returns false if the conn was closed. Handlers may close this conn (write
-  // error, closeSession, expired handshake) or any other conn (watch delivery,
-  // session steal), so the conn is re-resolved by id after every frame.
-  bool process_frames_locked(Conn* c) {
-    uint64_t cid = c->id;
-    while (true) {
+  // returns false if the conn was closed (dead conns are closed by caller)
+  bool process_frames(Conn* c) {
+    while (!c->dead.load()) {
       size_t avail = c->inbuf.size() - c->inpos;
       if (avail < 4) break;
       const unsigned char* p = reinterpret_cast<const unsigned char*>(c->inbuf.data() + c->inpos);
       uint32_t len = (static_cast<uint32_t>(p[0]) << 24) | (static_cast<uint32_t>(p[1]) << 16) |
                      (static_cast<uint32_t>(p[2]) << 8) | static_cast<uint32_t>(p[3]);
       if (len > 4 * 1024 * 1024) {  // jute.maxbuffer-ish sanity cap
-        close_conn_locked(c);
+        close_conn(c);
         return false;
       }
       if (avail < 4 + len) break;
-      // frame body lives in c->inbuf; handlers must not touch inbuf (they
-      // don't — they only parse `body` and enqueue output)
       const char* body = c->inbuf.data() + c->inpos + 4;
       c->inpos += 4 + len;
-      handle_frame_locked(c, body, len);
-      auto it = conns.find(cid);
-      if (it == conns.end()) return false;
-      c = it->second.get();
+      if (!handle_frame(c, body, len)) {
+        close_conn(c);
+        return false;
+      }
     }
     if (c->inpos > 0) {
       c->inbuf.erase(0, c->inpos);
@@ -288,101 +324,126 @@ struct Ensemble::Impl {
     return true;
   }
 
-  void handle_frame_locked(Conn* c, const char* body, size_t len) {
+  // returns false ⇒ caller closes the conn (protocol error)
+  bool handle_frame(Conn* c, const char* body, size_t len) {
     try {
       JuteReader r(body, len);
       if (!c->handshaken) {
-        handle_connect_locked(c, r);
-        return;
+        handle_connect(c, r);
+        return true;
       }
       RequestHeader hdr;
       hdr.deserialize(r);
-      touch_session_locked(c->session_id);
       switch (hdr.type) {
-        case kOpPing:
-          count_op("ping");
-          send_reply_locked(c, kXidPing, kZOk, nullptr);
+        case kOpPing: {
+          ops.ping++;
+          int64_t z;
+          {
+            std::lock_guard<std::mutex> g(mu);
+            touch_session_locked(c->session_id);
+            z = zxid_counter;
+          }
+          send_reply(c, kXidPing, z, kZOk, nullptr);
           break;
+        }
         case kOpCreate:
-          handle_create_locked(c, hdr.xid, r);
+          handle_create(c, hdr.xid, r);
           break;
         case kOpDelete:
-          handle_delete_locked(c, hdr.xid, r);
+          handle_delete(c, hdr.xid, r);
           break;
         case kOpExists:
-          handle_exists_locked(c, hdr.xid, r);
+          handle_exists(c, hdr.xid, r);
           break;
         case kOpGetData:
-          handle_get_data_locked(c, hdr.xid, r);
+          handle_get_data(c, hdr.xid, r);
           break;
         case kOpSetData:
-          handle_set_data_locked(c, hdr.xid, r);
+          handle_set_data(c, hdr.xid, r);
           break;
         case kOpGetChildren:
-          handle_get_children_locked(c, hdr.xid, r);
+          handle_get_children(c, hdr.xid, r);
           break;
         case kOpCloseSession:
-          handle_close_session_locked(c, hdr.xid);
-          return;
-        default:
-          count_op("unknown");
-          send_reply_locked(c, hdr.xid, kZSystemError, nullptr);
+          handle_close_session(c, hdr.xid);
           break;
+        default: {
+          ops.unknown++;
+          int64_t z;
+          {
+            std::lock_guard<std::mutex> g(mu);
+            touch_session_locked(c->session_id);
+            z = zxid_counter;
+          }
+          send_reply(c, hdr.xid, z, kZSystemError, nullptr);
+          break;
+        }
       }
+      return true;
     } catch (const std::exception& e) {
       log.warn("ensemble: malformed frame, closing conn", {{"err", Json(e.what())}});
-      if (conns.count(c->id)) close_conn_locked(c);
+      return false;
     }
   }
 
-  void handle_connect_locked(Conn* c, JuteReader& r) {
+  void handle_connect(Conn* c, JuteReader& r) {
     ConnectRequest req;
     req.deserialize(r);
-    count_op("connect");
+    ops.connect++;
     ConnectResponse resp;
     resp.has_read_only = req.has_read_only;
-    if (req.session_id != 0) {
-      auto sit = sessions.find(req.session_id);
-      if (sit == sessions.end() || sit->second.passwd != req.passwd) {
-        // unknown/expired/bad-passwd session ⇒ the canonical "expired"
-        // ConnectResponse: sessionId=0, timeOut=0
-        resp.session_id = 0;
-        resp.time_out_ms = 0;
-        log.info("connect: session expired/unknown", {{"session", Json(req.session_id)}});
-      } else {
-        Session& s = sit->second;
-        if (s.conn_id != 0) {
-          auto old = conns.find(s.conn_id);
-          if (old != conns.end()) close_conn_locked(old->second.get(), /*detach_session=*/false);
+    bool expired_handshake = false;
+    {
+      std::lock_guard<std::mutex> g(mu);
+      if (req.session_id != 0) {
+        auto sit = sessions.find(req.session_id);
+        if (sit == sessions.end() || sit->second.passwd != req.passwd) {
+          // unknown/expired/bad-passwd session ⇒ the canonical "expired"
+          // ConnectResponse: sessionId=0, timeOut=0
+          resp.session_id = 0;
+          resp.time_out_ms = 0;
+          expired_handshake = true;
+          log.info("connect: session expired/unknown", {{"session", Json(req.session_id)}});
+        } else {
+          Session& s = sit->second;
+          if (s.conn_id != 0) {
+            auto old = conns.find(s.conn_id);
+            if (old != conns.end()) {
+              // session moved: retire the old connection without detaching
+              old->second->dead.store(true);
+              post_close_locked(old->second);
+            }
+          }
+          s.conn_id = c->id;
+          s.last_touch = now_ms();
+          c->session_id = s.id;
+          resp.session_id = s.id;
+          resp.time_out_ms = s.timeout_ms;
+          resp.passwd = s.passwd;
+          log.info("connect: session re-attached", {{"session", Json(s.id)}});
         }
-        s.conn_id = c->id;
+      } else {
+        Session s;
+        s.id = next_session++;
+        s.passwd.resize(16);
+        uint64_t seed = static_cast<uint64_t>(s.id) * 0x9E3779B97F4A7C15ull + 0xD1B54A32D192ED03ull;
+        for (int i = 0; i < 16; i++) {
+          seed ^= seed >> 27;
+          seed *= 0x94D049BB133111EBull;
+          s.passwd[i] = static_cast<char>(seed >> (8 * (i % 8)));
+        }
+        int req_to = req.time_out_ms > 0 ? req.time_out_ms : 30000;
+        s.timeout_ms = std::max(cfg.min_session_timeout_ms, std::min(cfg.max_session_timeout_ms, req_to));
         s.last_touch = now_ms();
+        s.conn_id = c->id;
         c->session_id = s.id;
         resp.session_id = s.id;
         resp.time_out_ms = s.timeout_ms;
         resp.passwd = s.passwd;
-        log.info("connect: session re-attached", {{"session", Json(s.id)}});
+        sessions[s.id] = s;
+        log.info("connect: new session",
+                 {{"session", Json(s.id)}, {"timeout_ms", Json(static_cast<int64_t>(s.timeout_ms))}});
       }
-    } else {
-      Session s;
-      s.id = next_session++;
-      s.passwd.resize(16);
-      uint64_t seed = static_cast<uint64_t>(s.id) * 0x9E3779B97F4A7C15ull + 0xD1B54A32D192ED03ull;
-      for (int i = 0; i < 16; i++) {
-        seed ^= seed >> 27;
-        seed *= 0x94D049BB133111EBull;
-        s.passwd[i] = static_cast<char>(seed >> (8 * (i % 8)));
-      }
-      int req_to = req.time_out_ms > 0 ? req.time_out_ms : 30000;
-      s.timeout_ms = std::max(cfg.min_session_timeout_ms, std::min(cfg.max_session_timeout_ms, req_to));
-      s.last_touch = now_ms();
-      s.conn_id = c->id;
-      c->session_id = s.id;
-      resp.session_id = s.id;
-      resp.time_out_ms = s.timeout_ms;
-      resp.passwd = s.passwd;
-      sessions[s.id] = s;
-      log.info("connect: new session", {{"session", Json(s.id)}, {"timeout_ms", Json(static_cast<int64_t>(s.timeout_ms))}});
     }
     c->handshaken = true;
     std::string pkt;
@@ -390,87 +451,103 @@ struct Ensemble::Impl {
     JuteWriter w(&pkt);
     resp.serialize(w);
     frame_packet(&pkt);
-    send_raw_locked(c, std::move(pkt));
-    if (resp.session_id == 0) {
-      // expired handshake: server closes after notifying
-      shutdown_after_flush_locked(c);
+    if (expired_handshake) {
+      // expired handshake: server notifies then closes
+      {
+        std::lock_guard<std::mutex> og(c->out_mu);
+        c->closing = true;
+      }
+      send_packet(c, std::move(pkt));
+    } else {
+      send_packet(c, std::move(pkt));
     }
   }
 
   // --- ops ---
 
-  void handle_create_locked(Conn* c, int32_t xid, JuteReader& r) {
-    count_op("create");
+  void handle_create(Conn* c, int32_t xid, JuteReader& r) {
+    ops.create++;
     CreateRequest req;
     req.deserialize(r);
-    if (!valid_path(req.path) || req.path == "/") {
-      send_reply_locked(c, xid, kZMarshallingError, nullptr);
-      return;
-    }
-    std::string parent = parent_path(req.path);
-    auto pit = nodes.find(parent);
-    if (pit == nodes.end()) {
-      send_reply_locked(c, xid, kZNoNode, nullptr);
-      return;
-    }
-    if (pit->second.stat.ephemeral_owner != 0) {
-      send_reply_locked(c, xid, kZNoChildrenForEphemerals, nullptr);
-      return;
-    }
-    std::string path = req.path;
-    if (req.flags & kSequence) {
-      char suffix[16];
-      snprintf(suffix, sizeof(suffix), "%010d", pit->second.stat.cversion);
-      path += suffix;
-    }
-    if (nodes.count(path)) {
-      send_reply_locked(c, xid, kZNodeExists, nullptr);
-      return;
-    }
-    int64_t z = ++zxid_counter;
-    ZNode n;
-    n.data = req.data;
-    n.stat.czxid = z;
-    n.stat.mzxid = z;
-    n.stat.ctime = wall_ms();
-    n.stat.mtime = n.stat.ctime;
-    n.stat.data_length = static_cast<int32_t>(req.data.size());
-    if (req.flags & kEphemeral) {
-      n.stat.ephemeral_owner = c->session_id;
-      ephemerals[c->session_id].insert(path);
-    }
-    nodes[path] = std::move(n);
-    ZNode& par = nodes[parent];
-    par.children.insert(basename_of(path));
-    par.stat.cversion++;
-    par.stat.pzxid = z;
-    par.stat.num_children = static_cast<int32_t>(par.children.size());
-    fire_data_watches_locked(path, kEventNodeCreated);
-    fire_child_watches_locked(parent);
+    int32_t err = kZOk;
+    int64_t z;
     CreateResponse resp;
-    resp.path = path;
-    send_reply_locked(c, xid, kZOk, [&](JuteWriter& w) { resp.serialize(w); });
+    {
+      std::lock_guard<std::mutex> g(mu);
+      touch_session_locked(c->session_id);
+      if (!valid_path(req.path) || req.path == "/") {
+        err = kZMarshallingError;
+      } else {
+        std::string parent = parent_path(req.path);
+        auto pit = nodes.find(parent);
+        if (pit == nodes.end()) {
+          err = kZNoNode;
+        } else if (pit->second.stat.ephemeral_owner != 0) {
+          err = kZNoChildrenForEphemerals;
+        } else {
+          std::string path = req.path;
+          if (req.flags & kSequence) {
+            char suffix[16];
+            snprintf(suffix, sizeof(suffix), "%010d", pit->second.stat.cversion);
+            path += suffix;
+          }
+          if (nodes.count(path)) {
+            err = kZNodeExists;
+          } else {
+            int64_t zz = ++zxid_counter;
+            ZNode n;
+            n.data = req.data;
+            n.stat.czxid = zz;
+            n.stat.mzxid = zz;
+            n.stat.ctime = wall_ms();
+            n.stat.mtime = n.stat.ctime;
+            n.stat.data_length = static_cast<int32_t>(req.data.size());
+            if (req.flags & kEphemeral) {
+              n.stat.ephemeral_owner = c->session_id;
+              ephemerals[c->session_id].insert(path);
+            }
+            nodes[path] = std::move(n);
+            ZNode& par = nodes[parent];
+            par.children.insert(basename_of(path));
+            par.stat.cversion++;
+            par.stat.pzxid = zz;
+            par.stat.num_children = static_cast<int32_t>(par.children.size());
+            fire_data_watches_locked(path, kEventNodeCreated);
+            fire_child_watches_locked(parent);
+            resp.path = path;
+          }
+        }
+      }
+      z = zxid_counter;
+    }
+    if (err == kZOk)
+      send_reply(c, xid, z, kZOk, [&](JuteWriter& w) { resp.serialize(w); });
+    else
+      send_reply(c, xid, z, err, nullptr);
   }
 
-  void handle_delete_locked(Conn* c, int32_t xid, JuteReader& r) {
-    count_op("delete");
+  void handle_delete(Conn* c, int32_t xid, JuteReader& r) {
+    ops.del++;
     DeleteRequest req;
     req.deserialize(r);
-    auto it = nodes.find(req.path);
-    if (it == nodes.end() || req.path == "/") {
-      send_reply_locked(c, xid, kZNoNode, nullptr);
-      return;
+    int32_t err = kZOk;
+    int64_t z;
+    {
+      std::lock_guard<std::mutex> g(mu);
+      touch_session_locked(c->session_id);
+      auto it = nodes.find(req.path);
+      if (it == nodes.end() || req.path == "/") {
+        err = kZNoNode;
+      } else if (!it->second.children.empty()) {
+        err = kZNotEmpty;
+      } else if (req.version != -1 && req.version != it->second.stat.version) {
+        err = kZBadVersion;
+      } else {
+        delete_node_locked(req.path);
+      }
+      z = zxid_counter;
     }
-    if (!it->second.children.empty()) {
-      send_reply_locked(c, xid, kZNotEmpty, nullptr);
-      return;
-    }
-    if (req.version != -1 && req.version != it->second.stat.version) {
-      send_reply_locked(c, xid, kZBadVersion, nullptr);
-      return;
-    }
-    delete_node_locked(req.path);
-    send_reply_locked(c, xid, kZOk, nullptr);
+    send_reply(c, xid, z, err, nullptr);
   }
 
   // mu held; fires watches, updates parent
@@ -495,94 +572,144 @@ struct Ensemble::Impl {
     fire_data_watches_locked(path, kEventNodeDeleted);
   }
 
-  void handle_exists_locked(Conn* c, int32_t xid, JuteReader& r) {
-    count_op("exists");
+  void handle_exists(Conn* c, int32_t xid, JuteReader& r) {
+    ops.exists++;
     ExistsRequest req;
     req.deserialize(r);
-    auto it = nodes.find(req.path);
-    if (req.watch) data_watches[req.path].insert(c->session_id);
-    if (it == nodes.end()) {
-      send_reply_locked(c, xid, kZNoNode, nullptr);
-      return;
-    }
+    int32_t err = kZOk;
+    int64_t z;
     ExistsResponse resp;
-    resp.stat = it->second.stat;
-    send_reply_locked(c, xid, kZOk, [&](JuteWriter& w) { resp.serialize(w); });
+    {
+      std::lock_guard<std::mutex> g(mu);
+      touch_session_locked(c->session_id);
+      if (req.watch) data_watches[req.path].insert(c->session_id);
+      auto it = nodes.find(req.path);
+      if (it == nodes.end())
+        err = kZNoNode;
+      else
+        resp.stat = it->second.stat;
+      z = zxid_counter;
+    }
+    if (err == kZOk)
+      send_reply(c, xid, z, kZOk, [&](JuteWriter& w) { resp.serialize(w); });
+    else
+      send_reply(c, xid, z, err, nullptr);
   }
 
-  void handle_get_data_locked(Conn* c, int32_t xid, JuteReader& r) {
-    count_op("getData");
+  void handle_get_data(Conn* c, int32_t xid, JuteReader& r) {
+    ops.get_data++;
     GetDataRequest req;
     req.deserialize(r);
-    auto it = nodes.find(req.path);
-    if (it == nodes.end()) {
-      send_reply_locked(c, xid, kZNoNode, nullptr);
-      return;
-    }
-    if (req.watch) data_watches[req.path].insert(c->session_id);
+    int32_t err = kZOk;
+    int64_t z;
     GetDataResponse resp;
-    resp.data = it->second.data;
-    resp.stat = it->second.stat;
-    send_reply_locked(c, xid, kZOk, [&](JuteWriter& w) { resp.serialize(w); });
+    {
+      std::lock_guard<std::mutex> g(mu);
+      touch_session_locked(c->session_id);
+      auto it = nodes.find(req.path);
+      if (it == nodes.end()) {
+        err = kZNoNode;
+      } else {
+        if (req.watch) data_watches[req.path].insert(c->session_id);
+        resp.data = it->second.data;
+        resp.stat = it->second.stat;
+      }
+      z = zxid_counter;
+    }
+    if (err == kZOk)
+      send_reply(c, xid, z, kZOk, [&](JuteWriter& w) { resp.serialize(w); });
+    else
+      send_reply(c, xid, z, err, nullptr);
   }
 
-  void handle_set_data_locked(Conn* c, int32_t xid, JuteReader& r) {
-    count_op("setData");
+  void handle_set_data(Conn* c, int32_t xid, JuteReader& r) {
+    ops.set_data++;
     SetDataRequest req;
     req.deserialize(r);
-    auto it = nodes.find(req.path);
-    if (it == nodes.end()) {
-      send_reply_locked(c, xid, kZNoNode, nullptr);
-      return;
-    }
-    if (req.version != -1 && req.version != it->second.stat.version) {
-      send_reply_locked(c, xid, kZBadVersion, nullptr);
-      return;
-    }
-    int64_t z = ++zxid_counter;
-    it->second.data = req.data;
-    it->second.stat.mzxid = z;
-    it->second.stat.mtime = wall_ms();
-    it->second.stat.version++;
-    it->second.stat.data_length = static_cast<int32_t>(req.data.size());
-    fire_data_watches_locked(req.path, kEventNodeDataChanged);
+    int32_t err = kZOk;
+    int64_t z;
     SetDataResponse resp;
-    resp.stat = it->second.stat;
-    send_reply_locked(c, xid, kZOk, [&](JuteWriter& w) { resp.serialize(w); });
+    {
+      std::lock_guard<std::mutex> g(mu);
+      touch_session_locked(c->session_id);
+      auto it = nodes.find(req.path);
+      if (it == nodes.end()) {
+        err = kZNoNode;
+      } else if (req.version != -1 && req.version != it->second.stat.version) {
+        err = kZBadVersion;
+      } else {
+        int64_t zz = ++zxid_counter;
+        it->second.data = req.data;
+        it->second.stat.mzxid = zz;
+        it->second.stat.mtime = wall_ms();
+        it->second.stat.version++;
+        it->second.stat.data_length = static_cast<int32_t>(req.data.size());
+        fire_data_watches_locked(req.path, kEventNodeDataChanged);
+        resp.stat = it->second.stat;
+      }
+      z = zxid_counter;
+    }
+    if (err == kZOk)
+      send_reply(c, xid, z, kZOk, [&](JuteWriter& w) { resp.serialize(w); });
+    else
+      send_reply(c, xid, z, err, nullptr);
   }
 
-  void handle_get_children_locked(Conn* c, int32_t xid, JuteReader& r) {
-    count_op("getChildren");
+  void handle_get_children(Conn* c, int32_t xid, JuteReader& r) {
+    ops.get_children++;
     GetChildrenRequest req;
     req.deserialize(r);
-    auto it = nodes.find(req.path);
-    if (it == nodes.end()) {
-      send_reply_locked(c, xid, kZNoNode, nullptr);
-      return;
-    }
-    if (req.watch) child_watches[req.path].insert(c->session_id);
+    int32_t err = kZOk;
+    int64_t z;
     GetChildrenResponse resp;
-    resp.children.assign(it->second.children.begin(), it->second.children.end());
-    send_reply_locked(c, xid, kZOk, [&](JuteWriter& w) { resp.serialize(w); });
+    {
+      std::lock_guard<std::mutex> g(mu);
+      touch_session_locked(c->session_id);
+      auto it = nodes.find(req.path);
+      if (it == nodes.end()) {
+        err = kZNoNode;
+      } else {
+        if (req.watch) child_watches[req.path].insert(c->session_id);
+        resp.children.assign(it->second.children.begin(), it->second.children.end());
+      }
+      z = zxid_counter;
+    }
+    if (err == kZOk)
+      send_reply(c, xid, z, kZOk, [&](JuteWriter& w) { resp.serialize(w); });
+    else
+      send_reply(c, xid, z, err, nullptr);
   }
 
-  void handle_close_session_locked(Conn* c, int32_t xid) {
-    count_op("closeSession");
-    send_reply_locked(c, xid, kZOk, nullptr);
+  void handle_close_session(Conn* c, int32_t xid) {
+    ops.close_session++;
     int64_t sid = c->session_id;
-    shutdown_after_flush_locked(c);
-    if (sid != 0) kill_session_locked(sid, /*notify_conn=*/false);
+    int64_t z;
+    {
+      std::lock_guard<std::mutex> g(mu);
+      z = zxid_counter;
+    }
+    {
+      std::lock_guard<std::mutex> og(c->out_mu);
+      c->closing = true;  // close after the reply drains
+    }
+    send_reply(c, xid, z, kZOk, nullptr);
+    if (sid != 0) {
+      std::lock_guard<std::mutex> g(mu);
+      kill_session_locked(sid, /*close_conn=*/false);
+    }
   }
 
   // ---------------- session lifecycle ----------------
 
+  // mu held
   void touch_session_locked(int64_t sid) {
     auto it = sessions.find(sid);
     if (it != sessions.end()) it->second.last_touch = now_ms();
   }
 
-  // Expire/close a session: remove ephemerals (firing watches), tombstone it.
-  void kill_session_locked(int64_t sid, bool notify_conn) {
+  // mu held. Expire/close a session: remove ephemerals (firing watches),
+  // tombstone it, optionally retire its connection.
+  void kill_session_locked(int64_t sid, bool close_conn_too) {
     auto it = sessions.find(sid);
     if (it == sessions.end()) return;
     uint64_t cid = it->second.conn_id;
@@ -592,19 +719,21 @@ struct Ensemble::Impl {
       for (const auto& p : paths) delete_node_locked(p);
       ephemerals.erase(sid);
     }
-    // drop its watches
     for (auto& kv : data_watches) kv.second.erase(sid);
     for (auto& kv : child_watches) kv.second.erase(sid);
     sessions.erase(sid);
     dead_sessions.insert(sid);
-    if (notify_conn && cid != 0) {
+    if (close_conn_too && cid != 0) {
       auto cit = conns.find(cid);
-      if (cit != conns.end()) close_conn_locked(cit->second.get(), /*detach_session=*/false);
+      if (cit != conns.end()) {
+        cit->second->dead.store(true);
+        post_close_locked(cit->second);
+      }
     }
   }
 
   void schedule_sweep() {
-    loop.schedule(cfg.tick_ms, [this] {
+    loops[0]->schedule(cfg.tick_ms, [this] {
       {
         std::lock_guard<std::mutex> g(mu);
         int64_t now = now_ms();
@@ -613,7 +742,7 @@ struct Ensemble::Impl {
           if (now - kv.second.last_touch > kv.second.timeout_ms) expired.push_back(kv.first);
         for (int64_t sid : expired) {
           log.info("session expired", {{"session", Json(sid)}});
-          kill_session_locked(sid, /*notify_conn=*/true);
+          kill_session_locked(sid, /*close_conn=*/true);
         }
       }
       schedule_sweep();
@@ -622,6 +751,7 @@ struct Ensemble::Impl {
 
   // ---------------- watches ----------------
 
+  // mu held for all three
   void fire_data_watches_locked(const std::string& path, int32_t event_type) {
     auto it = data_watches.find(path);
     if (it == data_watches.end()) return;
@@ -658,53 +788,75 @@ struct Ensemble::Impl {
       ev.path = path;
       ev.serialize(w);
       frame_packet(&pkt);
-      send_raw_locked(cit->second.get(), std::move(pkt));
+      send_packet(cit->second.get(), std::move(pkt));
     }
   }
 
   // ---------------- response sending ----------------
 
-  void count_op(const char* name) { op_counters[name]++; }
-
   template <typename BodyFn>
-  void send_reply_locked(Conn* c, int32_t xid, int32_t err, BodyFn body) {
+  void send_reply(Conn* c, int32_t xid, int64_t zxid, int32_t err, BodyFn body) {
     std::string pkt;
     begin_packet(&pkt);
     JuteWriter w(&pkt);
     ReplyHeader hdr;
     hdr.xid = xid;
-    hdr.zxid = zxid_counter;
+    hdr.zxid = zxid;
     hdr.err = err;
     hdr.serialize(w);
     if constexpr (!std::is_same_v<BodyFn, std::nullptr_t>) {
       if (err == kZOk) body(w);
     }
     frame_packet(&pkt);
-    send_raw_locked(c, std::move(pkt));
+    send_packet(c, std::move(pkt));
   }
 
-  void send_reply_locked(Conn* c, int32_t xid, int32_t err, std::nullptr_t) {
-    send_reply_locked<std::nullptr_t>(c, xid, err, nullptr);
+  void send_reply(Conn* c, int32_t xid, int64_t zxid, int32_t err, std::nullptr_t) {
+    send_reply<std::nullptr_t>(c, xid, zxid, err, nullptr);
   }
 
-  void send_raw_locked(Conn* c, std::string pkt) {
+  // Any thread. Latency injection routes through the owner loop's timers
+  // (FIFO for equal deadlines), preserving per-conn response order.
+  void send_packet(Conn* c, std::string pkt) {
     int lat = latency_ms.load();
     if (lat > 0) {
       uint64_t cid = c->id;
-      loop.schedule(lat, [this, cid, pkt = std::move(pkt)]() mutable {
-        std::lock_guard<std::mutex> g(mu);
-        auto it = conns.find(cid);
-        if (it == conns.end()) return;
-        enqueue_locked(it->second.get(), std::move(pkt));
+      loops[c->server_idx]->schedule_from_any(lat, [this, cid, pkt = std::move(pkt)]() mutable {
+        ConnPtr cp = lookup(cid);
+        if (cp) deliver(cp.get(), std::move(pkt));
       });
       return;
     }
-    enqueue_locked(c, std::move(pkt));
+    if (loops[c->server_idx]->on_loop_thread()) {
+      // reply path: queue only; flushed once per event drain (on_conn_event)
+      // with a same-iteration 0 ms timer as the safety net for sends outside
+      // the event path (sweep-timer watch deliveries)
+      {
+        std::lock_guard<std::mutex> og(c->out_mu);
+        if (c->dead.load()) return;
+        c->outbuf += pkt;
+      }
+      if (!c->flush_scheduled) {
+        c->flush_scheduled = true;
+        uint64_t cid = c->id;
+        loops[c->server_idx]->schedule(0, [this, cid] {
+          ConnPtr cp = lookup(cid);
+          if (!cp) return;
+          cp->flush_scheduled = false;
+          if (!cp->dead.load()) flush_out(cp.get());
+        });
+      }
+      return;
+    }
+    deliver(c, std::move(pkt));
   }
 
-  void enqueue_locked(Conn* c, std::string pkt) {
+  // Any thread. Direct write when the buffer is clear, else buffer + arm
+  // EPOLLOUT on the owner loop.
+  void deliver(Conn* c, std::string pkt) {
+    std::lock_guard<std::mutex> og(c->out_mu);
+    if (c->dead.load()) return;
     if (c->outbuf.empty()) {
-      // fast path: try a direct write before buffering
       size_t off = 0;
       while (off < pkt.size()) {
         ssize_t n = write(c->fd, pkt.data() + off, pkt.size() - off);
@@ -715,103 +867,135 @@ struct Ensemble::Impl {
         } else if (n < 0 && errno == EINTR) {
           continue;
         } else {
-          close_conn_locked(c);
+          mark_dead(c);
           return;
         }
       }
       if (off < pkt.size()) {
         c->outbuf = pkt.substr(off);
-        loop.mod_fd(c->fd, EPOLLIN | EPOLLOUT);
+        arm_epollout(c);
       } else if (c->closing) {
-        close_conn_locked(c);
+        mark_dead(c);
       }
       return;
     }
     c->outbuf += pkt;
   }
 
-  void flush_out_locked(Conn* c) {
-    size_t off = 0;
-    while (off < c->outbuf.size()) {
-      ssize_t n = write(c->fd, c->outbuf.data() + off, c->outbuf.size() - off);
-      if (n > 0) {
-        off += static_cast<size_t>(n);
-      } else if (n < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) {
-        break;
-      } else if (n < 0 && errno == EINTR) {
-        continue;
-      } else {
-        close_conn_locked(c);
-        return;
-      }
-    }
-    c->outbuf.erase(0, off);
-    if (c->outbuf.empty()) {
-      if (c->closing) {
-        close_conn_locked(c);
-        return;
-      }
-      loop.mod_fd(c->fd, EPOLLIN);
+  // out_mu held
+  void arm_epollout(Conn* c) {
+    if (!c->epollout_armed) {
+      c->epollout_armed = true;
+      loops[c->server_idx]->mod_fd(c->fd, EPOLLIN | EPOLLOUT);
     }
   }
 
-  void shutdown_after_flush_locked(Conn* c) {
-    if (c->outbuf.empty()) {
-      close_conn_locked(c);
-    } else {
-      c->closing = true;
-    }
+  // out_mu held (or unambiguous owner). Marks the conn for closure; the owner
+  // loop performs the actual close.
+  void mark_dead(Conn* c) {
+    if (c->dead.exchange(true)) return;
+    uint64_t cid = c->id;
+    loops[c->server_idx]->post([this, cid] {
+      ConnPtr cp = lookup(cid);
+      if (cp) close_conn(cp.get());
+    });
   }
 
-  void close_conn_locked(Conn* c, bool detach_session = true) {
-    if (detach_session && c->session_id != 0) {
+  void flush_out(Conn* c) {
+    bool close_now = false;
+    {
+      std::lock_guard<std::mutex> og(c->out_mu);
+      size_t off = 0;
+      while (off < c->outbuf.size()) {
+        ssize_t n = write(c->fd, c->outbuf.data() + off, c->outbuf.size() - off);
+        if (n > 0) {
+          off += static_cast<size_t>(n);
+        } else if (n < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) {
+          break;
+        } else if (n < 0 && errno == EINTR) {
+          continue;
+        } else {
+          close_now = true;
+          break;
+        }
+      }
+      if (!close_now) {
+        c->outbuf.erase(0, off);
+        if (c->outbuf.empty()) {
+          c->epollout_armed = false;
+          loops[c->server_idx]->mod_fd(c->fd, EPOLLIN);
+          if (c->closing) close_now = true;
+        }
+      }
+    }
+    if (close_now) close_conn(c);
+  }
+
+  // MUST run on the owner loop thread.
+  void close_conn(Conn* c) {
+    c->dead.store(true);
+    std::lock_guard<std::mutex> g(mu);
+    auto it = conns.find(c->id);
+    if (it == conns.end()) return;  // already closed
+    if (c->session_id != 0) {
       auto sit = sessions.find(c->session_id);
       if (sit != sessions.end() && sit->second.conn_id == c->id) sit->second.conn_id = 0;
     }
-    loop.del_fd(c->fd);
-    close(c->fd);
-    conns.erase(c->id);
+    loops[c->server_idx]->del_fd(c->fd);
+    ::close(c->fd);
+    conns.erase(it);
+  }
+
+  // mu held: queue a close on the conn's owner loop
+  void post_close_locked(const ConnPtr& c) {
+    uint64_t cid = c->id;
+    size_t srv = c->server_idx;
+    loops[srv]->post([this, cid] {
+      ConnPtr cp = lookup(cid);
+      if (cp) close_conn(cp.get());
+    });
   }
 
   // ---------------- control (any thread) ----------------
 
-  void run_on_loop(std::function<void()> fn) {
-    if (loop.on_loop_thread()) {
-      fn();
-      return;
-    }
+  void kill_server(size_t idx) {
+    if (idx >= loops.size()) return;
     std::promise<void> done;
-    loop.post([&] {
-      fn();
+    loops[idx]->post([this, idx, &done] {
+      std::vector<ConnPtr> victims;
+      {
+        std::lock_guard<std::mutex> g(mu);
+        if (idx < servers.size() && servers[idx].up) {
+          Server& s = servers[idx];
+          loops[idx]->del_fd(s.listen_fd);
+          ::close(s.listen_fd);
+          s.listen_fd = -1;
+          s.up = false;
+          for (auto& kv : conns)
+            if (kv.second->server_idx == idx) victims.push_back(kv.second);
+        }
+      }
+      for (auto& c : victims) close_conn(c.get());
+      log.info("server killed", {{"server", Json(static_cast<int64_t>(idx))}});
       done.set_value();
     });
     done.get_future().wait();
   }
 
-  void kill_server(size_t idx) {
-    run_on_loop([this, idx] {
-      std::lock_guard<std::mutex> g(mu);
-      if (idx >= servers.size() || !servers[idx].up) return;
-      Server& s = servers[idx];
-      loop.del_fd(s.listen_fd);
-      close(s.listen_fd);
-      s.listen_fd = -1;
-      s.up = false;
-      std::vector<Conn*> victims;
-      for (auto& kv : conns)
-        if (kv.second->server_idx == idx) victims.push_back(kv.second.get());
-      for (Conn* c : victims) close_conn_locked(c);
-      log.info("server killed", {{"server", Json(static_cast<int64_t>(idx))}});
-    });
-  }
-
   void restart_server(size_t idx) {
-    run_on_loop([this, idx] {
-      std::lock_guard<std::mutex> g(mu);
-      if (idx >= servers.size() || servers[idx].up) return;
-      open_listener(idx, servers[idx].port);
-      log.info("server restarted", {{"server", Json(static_cast<int64_t>(idx))}});
+    if (idx >= loops.size()) return;
+    std::promise<void> done;
+    loops[idx]->post([this, idx, &done] {
+      {
+        std::lock_guard<std::mutex> g(mu);
+        if (idx < servers.size() && !servers[idx].up) {
+          open_listener(idx, servers[idx].port);
+          log.info("server restarted", {{"server", Json(static_cast<int64_t>(idx))}});
+        }
+      }
+      done.set_value();
     });
+    done.get_future().wait();
   }
 
   // mu held
@@ -837,6 +1021,11 @@ struct Ensemble::Impl {
     if (cfg.election_ms > 0) election_until = now_ms() + cfg.election_ms;
     maybe_elect_leader_locked();
     return victim;
+  }
+
+  void expire_session(int64_t session_id) {
+    std::lock_guard<std::mutex> g(mu);
+    kill_session_locked(session_id, /*close_conn=*/true);
   }
 };
 
@@ -885,12 +1074,7 @@ size_t Ensemble::leader() const {
 
 size_t Ensemble::kill_leader() { return impl_->kill_leader(); }
 
-void Ensemble::expire_session(int64_t session_id) {
-  impl_->run_on_loop([this, session_id] {
-    std::lock_guard<std::mutex> g(impl_->mu);
-    impl_->kill_session_locked(session_id, /*notify_conn=*/true);
-  });
-}
+void Ensemble::expire_session(int64_t session_id) { impl_->expire_session(session_id); }
 
 void Ensemble::set_latency_ms(int ms) { impl_->latency_ms.store(ms); }
 
@@ -938,8 +1122,26 @@ int64_t Ensemble::zxid() const {
 }
 
 std::map<std::string, uint64_t> Ensemble::counters() const {
-  std::lock_guard<std::mutex> g(impl_->mu);
-  return impl_->op_counters;
+  std::map<std::string, uint64_t> out;
+  const auto& o = impl_->ops;
+  out["connect"] = o.connect.load();
+  out["create"] = o.create.load();
+  out["delete"] = o.del.load();
+  out["exists"] = o.exists.load();
+  out["getData"] = o.get_data.load();
+  out["setData"] = o.set_data.load();
+  out["getChildren"] = o.get_children.load();
+  out["ping"] = o.ping.load();
+  out["closeSession"] = o.close_session.load();
+  out["unknown"] = o.unknown.load();
+  // drop zero counters so tests can assert presence meaningfully
+  for (auto it = out.begin(); it != out.end();) {
+    if (it->second == 0)
+      it = out.erase(it);
+    else
+      ++it;
+  }
+  return out;
 }
 
 }  // namespace zk

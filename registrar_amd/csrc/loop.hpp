@@ -75,12 +75,14 @@ class EventLoop {
     fd_cbs_[fd] = std::move(cb);
   }
 
+  // Error-tolerant: the fd may already be closed/removed by another thread's
+  // shutdown path; epoll_ctl itself is thread-safe.
   void mod_fd(int fd, uint32_t events) {
     struct epoll_event ev;
     memset(&ev, 0, sizeof(ev));
     ev.events = events;
     ev.data.fd = fd;
-    if (epoll_ctl(epfd_, EPOLL_CTL_MOD, fd, &ev) < 0) throw std::runtime_error("epoll_ctl MOD failed");
+    epoll_ctl(epfd_, EPOLL_CTL_MOD, fd, &ev);
   }
 
   void del_fd(int fd) {
@@ -112,6 +114,15 @@ class EventLoop {
 
   void stop() {
     post([this] { running_ = false; });
+  }
+
+  // Timer arming from any thread (the heap itself is loop-thread-only).
+  void schedule_from_any(int64_t delay_ms, std::function<void()> fn) {
+    if (on_loop_thread()) {
+      schedule(delay_ms, std::move(fn));
+    } else {
+      post([this, delay_ms, fn = std::move(fn)]() mutable { schedule(delay_ms, std::move(fn)); });
+    }
   }
 
   // Run the loop until stop(). Call from the owning thread.
@@ -197,7 +208,11 @@ class EventLoop {
   struct HeapEntry {
     int64_t deadline;
     TimerId id;
-    bool operator>(const HeapEntry& o) const { return deadline > o.deadline; }
+    // FIFO among equal deadlines (ids are monotonic) so delayed responses
+    // keep their submission order (latency-injection correctness)
+    bool operator>(const HeapEntry& o) const {
+      return deadline != o.deadline ? deadline > o.deadline : id > o.id;
+    }
   };
 
   int epfd_ = -1;

@@ -97,6 +97,11 @@ struct ZkClient::Impl {
 
   Impl(ZkClientConfig c, Logger l) : cfg(std::move(c)), log(l.child("zookeeper")) {
     log.set_level(cfg.log_level);
+    // spread clients across the ensemble: random starting server (real ZK
+    // clients shuffle the connect string for the same reason)
+    server_idx = static_cast<size_t>((static_cast<uint64_t>(getpid()) * 0x9E3779B97F4A7C15ull +
+                                      reinterpret_cast<uintptr_t>(this)) >>
+                                     17);
     connect_backoff.initial_ms = cfg.connect_initial_delay_ms;
     connect_backoff.max_ms = cfg.connect_max_delay_ms;
     connect_backoff.max_attempts = cfg.connect_max_attempts;
