@@ -73,7 +73,7 @@ def main():
     if distributed:
         import torch.distributed as dist  # noqa: F811
 
-        backend = "nccl" if use_cuda else "gloo"
+        backend = os.environ.get("BENCH_BACKEND") or ("nccl" if use_cuda else "gloo")
         dist.init_process_group(backend=backend, rank=rank, world_size=world)
 
     def barrier():

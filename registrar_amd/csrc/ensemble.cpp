@@ -1,12 +1,21 @@
 // ensemble.cpp — synthetic in-process ZooKeeper ensemble (see ensemble.hpp).
 //
-// Threading model (the scaling core): one epoll loop THREAD PER SERVER, so N
-// registrar processes spread across the ensemble's servers get parallel
-// socket I/O and frame processing. The shared data tree / session table /
-// watch maps live under one state mutex (`mu`) whose critical sections are
-// memory-only (no syscalls); per-connection output buffers have their own
-// mutex so reply writes and cross-thread watch deliveries never serialize on
-// the global lock. Lock order: mu → conn->out_mu (never the reverse).
+// Threading model (the scaling core of the whole framework): one epoll loop
+// THREAD PER SERVER for parallel socket I/O, and SHARDED state so concurrent
+// registrar processes (which write disjoint domain subtrees) never contend:
+//
+//   - znodes + their watch maps live in 64 hash shards, each under its own
+//     mutex; an op locks only the shard(s) of the paths it touches (create/
+//     delete lock child+parent shards in index order),
+//   - sessions live under one session mutex, but per-request touch is a
+//     lock-free atomic store through the connection's cached session pointer,
+//   - zxid is a global atomic counter,
+//   - per-connection output buffers have their own mutex; replies queue per
+//     event drain and flush as a few large writes.
+//
+// Global lock order: shard(s) → session_mu → conns_mu → conn->out_mu.
+// kill_session drains the ephemeral set with no other lock held before
+// taking shard locks (see the phase comments).
 #include "ensemble.hpp"
 
 #include <arpa/inet.h>
@@ -63,19 +72,31 @@ bool valid_path(const std::string& p) {
 }  // namespace
 
 struct Ensemble::Impl {
+  static constexpr size_t kShards = 64;
+
   struct ZNode {
     std::string data;
     Stat stat;
     std::set<std::string> children;
   };
 
+  struct Shard {
+    std::mutex mu;
+    std::unordered_map<std::string, ZNode> nodes;
+    std::unordered_map<std::string, std::set<int64_t>> data_watches;
+    std::unordered_map<std::string, std::set<int64_t>> child_watches;
+  };
+
   struct Session {
     int64_t id = 0;
     std::string passwd;
     int timeout_ms = 30000;
-    int64_t last_touch = 0;  // monotonic ms
-    uint64_t conn_id = 0;    // 0 = detached
+    std::atomic<int64_t> last_touch{0};  // monotonic ms, lock-free touch
+    std::atomic<uint64_t> conn_id{0};    // 0 = detached
+    std::mutex eph_mu;
+    std::set<std::string> ephemerals;
   };
+  using SessionPtr = std::shared_ptr<Session>;
 
   struct Conn {
     uint64_t id = 0;
@@ -85,14 +106,14 @@ struct Ensemble::Impl {
     std::string inbuf;
     size_t inpos = 0;
     bool handshaken = false;
-    int64_t session_id = 0;
+    SessionPtr session;  // cached after handshake (lock-free touch)
+    bool flush_scheduled = false;
     // shared output state:
     std::mutex out_mu;
     std::string outbuf;
     bool epollout_armed = false;
     bool closing = false;  // close once outbuf drains
     std::atomic<bool> dead{false};
-    bool flush_scheduled = false;  // owner loop thread only
   };
   using ConnPtr = std::shared_ptr<Conn>;
 
@@ -113,29 +134,51 @@ struct Ensemble::Impl {
   std::vector<std::thread> threads;
   std::atomic<bool> started{false};
   std::atomic<int> latency_ms{0};
+  std::atomic<int64_t> zxid_counter{0};
   OpCounters ops;
 
-  // ---- state under mu ----
-  mutable std::mutex mu;
-  std::vector<Server> servers;
-  std::unordered_map<uint64_t, ConnPtr> conns;
-  uint64_t next_conn_id = 1;
-  std::unordered_map<std::string, ZNode> nodes;
-  std::unordered_map<int64_t, Session> sessions;
+  std::array<Shard, kShards> shards;
+
+  mutable std::mutex session_mu;
+  std::unordered_map<int64_t, SessionPtr> sessions;
   std::unordered_set<int64_t> dead_sessions;  // expired or closed: reconnect ⇒ expired
-  std::unordered_map<int64_t, std::set<std::string>> ephemerals;
-  std::unordered_map<std::string, std::set<int64_t>> data_watches;
-  std::unordered_map<std::string, std::set<int64_t>> child_watches;
-  int64_t zxid_counter = 0;
   int64_t next_session = 0x100000;
+
+  mutable std::mutex conns_mu;
+  std::unordered_map<uint64_t, ConnPtr> conns;
+  std::atomic<uint64_t> next_conn_id{1};
+
+  mutable std::mutex admin_mu;  // servers / leader / election
+  std::vector<Server> servers;
   size_t leader_idx = 0;
   int64_t election_until = 0;  // monotonic ms; connects refused until then
 
   explicit Impl(EnsembleConfig c) : cfg(std::move(c)), log(Logger("zk-ensemble").child("ensemble")) {
     log.set_level(cfg.log_level);
     latency_ms.store(cfg.latency_ms);
-    nodes["/"] = ZNode{};
+    shard_of("/").nodes["/"] = ZNode{};
   }
+
+  Shard& shard_of(const std::string& path) {
+    return shards[std::hash<std::string>{}(path) % kShards];
+  }
+  size_t shard_idx(const std::string& path) const {
+    return std::hash<std::string>{}(path) % kShards;
+  }
+
+  // Lock the shards of two paths without deadlock (index order; may be same).
+  struct TwoShardLock {
+    std::unique_lock<std::mutex> a, b;
+    TwoShardLock(Impl& impl, const std::string& p1, const std::string& p2) {
+      size_t i1 = impl.shard_idx(p1), i2 = impl.shard_idx(p2);
+      if (i1 == i2) {
+        a = std::unique_lock<std::mutex>(impl.shards[i1].mu);
+      } else {
+        a = std::unique_lock<std::mutex>(impl.shards[std::min(i1, i2)].mu);
+        b = std::unique_lock<std::mutex>(impl.shards[std::max(i1, i2)].mu);
+      }
+    }
+  };
 
   // ---------------- lifecycle ----------------
 
@@ -145,7 +188,7 @@ struct Ensemble::Impl {
     loops.clear();
     for (size_t i = 0; i < n; i++) loops.push_back(std::make_unique<EventLoop>());
     {
-      std::lock_guard<std::mutex> g(mu);
+      std::lock_guard<std::mutex> g(admin_mu);
       servers.resize(n);
       for (size_t i = 0; i < n; i++) open_listener(i, cfg.ports[i]);
     }
@@ -159,19 +202,18 @@ struct Ensemble::Impl {
 
   void stop() {
     if (!started.load()) return;
-    // close everything from each owner loop, then stop the loops
     std::vector<std::promise<void>> done(loops.size());
     for (size_t i = 0; i < loops.size(); i++) {
       loops[i]->post([this, i, &done] {
         std::vector<ConnPtr> victims;
         {
-          std::lock_guard<std::mutex> g(mu);
+          std::lock_guard<std::mutex> g(conns_mu);
           for (auto& kv : conns)
             if (kv.second->server_idx == i) victims.push_back(kv.second);
         }
         for (auto& c : victims) close_conn(c.get());
         {
-          std::lock_guard<std::mutex> g(mu);
+          std::lock_guard<std::mutex> g(admin_mu);
           if (i < servers.size() && servers[i].listen_fd >= 0) {
             loops[i]->del_fd(servers[i].listen_fd);
             ::close(servers[i].listen_fd);
@@ -189,7 +231,7 @@ struct Ensemble::Impl {
     started.store(false);
   }
 
-  // mu held; listener registration goes to loops[idx]
+  // admin_mu held
   void open_listener(size_t idx, int port) {
     int fd = socket(AF_INET, SOCK_STREAM | SOCK_CLOEXEC, 0);
     if (fd < 0) throw std::runtime_error("ensemble: socket() failed");
@@ -234,27 +276,30 @@ struct Ensemble::Impl {
       int fd = accept4(listen_fd, nullptr, nullptr, SOCK_NONBLOCK | SOCK_CLOEXEC);
       if (fd < 0) break;
       {
-        std::lock_guard<std::mutex> g(mu);
+        std::lock_guard<std::mutex> g(admin_mu);
         if (now_ms() < election_until) {
           // mid-election: nobody serves (BASELINE config 4 storm realism)
           ::close(fd);
           continue;
         }
         maybe_elect_leader_locked();
-        set_nodelay(fd);
-        auto conn = std::make_shared<Conn>();
-        conn->id = next_conn_id++;
-        conn->fd = fd;
-        conn->server_idx = server_idx;
-        conns[conn->id] = conn;
-        uint64_t cid = conn->id;
-        loops[server_idx]->add_fd(fd, EPOLLIN, [this, cid](uint32_t ev) { on_conn_event(cid, ev); });
       }
+      set_nodelay(fd);
+      auto conn = std::make_shared<Conn>();
+      conn->id = next_conn_id.fetch_add(1);
+      conn->fd = fd;
+      conn->server_idx = server_idx;
+      uint64_t cid = conn->id;
+      {
+        std::lock_guard<std::mutex> g(conns_mu);
+        conns[cid] = conn;
+      }
+      loops[server_idx]->add_fd(fd, EPOLLIN, [this, cid](uint32_t ev) { on_conn_event(cid, ev); });
     }
   }
 
   ConnPtr lookup(uint64_t cid) {
-    std::lock_guard<std::mutex> g(mu);
+    std::lock_guard<std::mutex> g(conns_mu);
     auto it = conns.find(cid);
     return it == conns.end() ? nullptr : it->second;
   }
@@ -297,7 +342,7 @@ struct Ensemble::Impl {
     if (ev & EPOLLOUT) flush_out(c.get());
   }
 
-  // returns false if the conn was closed (dead conns are closed by caller)
+  // returns false if the conn was closed
   bool process_frames(Conn* c) {
     while (!c->dead.load()) {
       size_t avail = c->inbuf.size() - c->inpos;
@@ -324,6 +369,10 @@ struct Ensemble::Impl {
     return true;
   }
 
+  void touch(Conn* c) {
+    if (c->session) c->session->last_touch.store(now_ms(), std::memory_order_relaxed);
+  }
+
   // returns false ⇒ caller closes the conn (protocol error)
   bool handle_frame(Conn* c, const char* body, size_t len) {
     try {
@@ -334,18 +383,12 @@ struct Ensemble::Impl {
       }
       RequestHeader hdr;
       hdr.deserialize(r);
+      touch(c);
       switch (hdr.type) {
-        case kOpPing: {
-          ops.ping++;
-          int64_t z;
-          {
-            std::lock_guard<std::mutex> g(mu);
-            touch_session_locked(c->session_id);
-            z = zxid_counter;
-          }
-          send_reply(c, kXidPing, z, kZOk, nullptr);
+        case kOpPing:
+          ops.ping.fetch_add(1, std::memory_order_relaxed);
+          send_reply(c, kXidPing, zxid_counter.load(std::memory_order_relaxed), kZOk, nullptr);
           break;
-        }
         case kOpCreate:
           handle_create(c, hdr.xid, r);
           break;
@@ -367,17 +410,10 @@ struct Ensemble::Impl {
         case kOpCloseSession:
           handle_close_session(c, hdr.xid);
           break;
-        default: {
-          ops.unknown++;
-          int64_t z;
-          {
-            std::lock_guard<std::mutex> g(mu);
-            touch_session_locked(c->session_id);
-            z = zxid_counter;
-          }
-          send_reply(c, hdr.xid, z, kZSystemError, nullptr);
+        default:
+          ops.unknown.fetch_add(1, std::memory_order_relaxed);
+          send_reply(c, hdr.xid, zxid_counter.load(std::memory_order_relaxed), kZSystemError, nullptr);
           break;
-        }
       }
       return true;
     } catch (const std::exception& e) {
@@ -389,15 +425,16 @@ struct Ensemble::Impl {
   void handle_connect(Conn* c, JuteReader& r) {
     ConnectRequest req;
     req.deserialize(r);
-    ops.connect++;
+    ops.connect.fetch_add(1, std::memory_order_relaxed);
     ConnectResponse resp;
     resp.has_read_only = req.has_read_only;
     bool expired_handshake = false;
+    ConnPtr old_conn;
     {
-      std::lock_guard<std::mutex> g(mu);
+      std::lock_guard<std::mutex> g(session_mu);
       if (req.session_id != 0) {
         auto sit = sessions.find(req.session_id);
-        if (sit == sessions.end() || sit->second.passwd != req.passwd) {
+        if (sit == sessions.end() || sit->second->passwd != req.passwd) {
           // unknown/expired/bad-passwd session ⇒ the canonical "expired"
           // ConnectResponse: sessionId=0, timeOut=0
           resp.session_id = 0;
@@ -405,45 +442,47 @@ struct Ensemble::Impl {
           expired_handshake = true;
           log.info("connect: session expired/unknown", {{"session", Json(req.session_id)}});
         } else {
-          Session& s = sit->second;
-          if (s.conn_id != 0) {
-            auto old = conns.find(s.conn_id);
-            if (old != conns.end()) {
-              // session moved: retire the old connection without detaching
-              old->second->dead.store(true);
-              post_close_locked(old->second);
-            }
+          SessionPtr s = sit->second;
+          uint64_t prev = s->conn_id.exchange(c->id);
+          if (prev != 0) {
+            std::lock_guard<std::mutex> cg(conns_mu);
+            auto old = conns.find(prev);
+            if (old != conns.end()) old_conn = old->second;
           }
-          s.conn_id = c->id;
-          s.last_touch = now_ms();
-          c->session_id = s.id;
-          resp.session_id = s.id;
-          resp.time_out_ms = s.timeout_ms;
-          resp.passwd = s.passwd;
-          log.info("connect: session re-attached", {{"session", Json(s.id)}});
+          s->last_touch.store(now_ms());
+          c->session = s;
+          resp.session_id = s->id;
+          resp.time_out_ms = s->timeout_ms;
+          resp.passwd = s->passwd;
+          log.info("connect: session re-attached", {{"session", Json(s->id)}});
         }
       } else {
-        Session s;
-        s.id = next_session++;
-        s.passwd.resize(16);
-        uint64_t seed = static_cast<uint64_t>(s.id) * 0x9E3779B97F4A7C15ull + 0xD1B54A32D192ED03ull;
+        auto s = std::make_shared<Session>();
+        s->id = next_session++;
+        s->passwd.resize(16);
+        uint64_t seed = static_cast<uint64_t>(s->id) * 0x9E3779B97F4A7C15ull + 0xD1B54A32D192ED03ull;
         for (int i = 0; i < 16; i++) {
           seed ^= seed >> 27;
           seed *= 0x94D049BB133111EBull;
-          s.passwd[i] = static_cast<char>(seed >> (8 * (i % 8)));
+          s->passwd[i] = static_cast<char>(seed >> (8 * (i % 8)));
         }
         int req_to = req.time_out_ms > 0 ? req.time_out_ms : 30000;
-        s.timeout_ms = std::max(cfg.min_session_timeout_ms, std::min(cfg.max_session_timeout_ms, req_to));
-        s.last_touch = now_ms();
-        s.conn_id = c->id;
-        c->session_id = s.id;
-        resp.session_id = s.id;
-        resp.time_out_ms = s.timeout_ms;
-        resp.passwd = s.passwd;
-        sessions[s.id] = s;
+        s->timeout_ms = std::max(cfg.min_session_timeout_ms, std::min(cfg.max_session_timeout_ms, req_to));
+        s->last_touch.store(now_ms());
+        s->conn_id.store(c->id);
+        c->session = s;
+        resp.session_id = s->id;
+        resp.time_out_ms = s->timeout_ms;
+        resp.passwd = s->passwd;
+        sessions[s->id] = s;
         log.info("connect: new session",
-                 {{"session", Json(s.id)}, {"timeout_ms", Json(static_cast<int64_t>(s.timeout_ms))}});
+                 {{"session", Json(s->id)}, {"timeout_ms", Json(static_cast<int64_t>(s->timeout_ms))}});
       }
+    }
+    if (old_conn) {
+      // session moved: retire the old connection without detaching
+      old_conn->dead.store(true);
+      post_close(old_conn);
     }
     c->handshaken = true;
     std::string pkt;
@@ -452,328 +491,366 @@ struct Ensemble::Impl {
     resp.serialize(w);
     frame_packet(&pkt);
     if (expired_handshake) {
-      // expired handshake: server notifies then closes
-      {
-        std::lock_guard<std::mutex> og(c->out_mu);
-        c->closing = true;
-      }
-      send_packet(c, std::move(pkt));
-    } else {
-      send_packet(c, std::move(pkt));
+      std::lock_guard<std::mutex> og(c->out_mu);
+      c->closing = true;  // expired handshake: notify, then close
     }
+    send_packet(c, std::move(pkt));
   }
+
+  int64_t sid_of(Conn* c) const { return c->session ? c->session->id : 0; }
 
   // --- ops ---
 
   void handle_create(Conn* c, int32_t xid, JuteReader& r) {
-    ops.create++;
+    ops.create.fetch_add(1, std::memory_order_relaxed);
     CreateRequest req;
     req.deserialize(r);
     int32_t err = kZOk;
-    int64_t z;
     CreateResponse resp;
-    {
-      std::lock_guard<std::mutex> g(mu);
-      touch_session_locked(c->session_id);
-      if (!valid_path(req.path) || req.path == "/") {
-        err = kZMarshallingError;
+    int64_t sid = sid_of(c);
+    bool made_ephemeral = false;
+    std::string created_path;
+    std::string child_watch_parent;
+    if (!valid_path(req.path) || req.path == "/") {
+      err = kZMarshallingError;
+    } else {
+      std::string parent = parent_path(req.path);
+      TwoShardLock lk(*this, req.path, parent);
+      Shard& psh = shard_of(parent);
+      auto pit = psh.nodes.find(parent);
+      if (pit == psh.nodes.end()) {
+        err = kZNoNode;
+      } else if (pit->second.stat.ephemeral_owner != 0) {
+        err = kZNoChildrenForEphemerals;
       } else {
-        std::string parent = parent_path(req.path);
-        auto pit = nodes.find(parent);
-        if (pit == nodes.end()) {
-          err = kZNoNode;
-        } else if (pit->second.stat.ephemeral_owner != 0) {
-          err = kZNoChildrenForEphemerals;
+        std::string path = req.path;
+        if (req.flags & kSequence) {
+          char suffix[16];
+          snprintf(suffix, sizeof(suffix), "%010d", pit->second.stat.cversion);
+          path += suffix;
+        }
+        // NB: with SEQUENCE the final path may hash to a different shard than
+        // req.path; re-lock correctly in that (registrar-unused) corner
+        Shard& csh = shard_of(path);
+        bool same_lock = (&csh == &shard_of(req.path)) || (&csh == &psh);
+        std::unique_lock<std::mutex> extra;
+        if (!same_lock) extra = std::unique_lock<std::mutex>(csh.mu, std::try_to_lock);
+        if (!same_lock && !extra.owns_lock()) {
+          err = kZSystemError;  // pathological shard collision; not reachable
+                                // for non-sequence creates
+        } else if (csh.nodes.count(path)) {
+          err = kZNodeExists;
         } else {
-          std::string path = req.path;
-          if (req.flags & kSequence) {
-            char suffix[16];
-            snprintf(suffix, sizeof(suffix), "%010d", pit->second.stat.cversion);
-            path += suffix;
+          int64_t zz = zxid_counter.fetch_add(1, std::memory_order_relaxed) + 1;
+          ZNode n;
+          n.data = req.data;
+          n.stat.czxid = zz;
+          n.stat.mzxid = zz;
+          n.stat.ctime = wall_ms();
+          n.stat.mtime = n.stat.ctime;
+          n.stat.data_length = static_cast<int32_t>(req.data.size());
+          if (req.flags & kEphemeral) {
+            n.stat.ephemeral_owner = sid;
+            made_ephemeral = true;
           }
-          if (nodes.count(path)) {
-            err = kZNodeExists;
-          } else {
-            int64_t zz = ++zxid_counter;
-            ZNode n;
-            n.data = req.data;
-            n.stat.czxid = zz;
-            n.stat.mzxid = zz;
-            n.stat.ctime = wall_ms();
-            n.stat.mtime = n.stat.ctime;
-            n.stat.data_length = static_cast<int32_t>(req.data.size());
-            if (req.flags & kEphemeral) {
-              n.stat.ephemeral_owner = c->session_id;
-              ephemerals[c->session_id].insert(path);
-            }
-            nodes[path] = std::move(n);
-            ZNode& par = nodes[parent];
-            par.children.insert(basename_of(path));
-            par.stat.cversion++;
-            par.stat.pzxid = zz;
-            par.stat.num_children = static_cast<int32_t>(par.children.size());
-            fire_data_watches_locked(path, kEventNodeCreated);
-            fire_child_watches_locked(parent);
-            resp.path = path;
-          }
+          csh.nodes[path] = std::move(n);
+          ZNode& par = pit->second;
+          par.children.insert(basename_of(path));
+          par.stat.cversion++;
+          par.stat.pzxid = zz;
+          par.stat.num_children = static_cast<int32_t>(par.children.size());
+          fire_data_watches_locked(csh, path, kEventNodeCreated);
+          fire_child_watches_locked(psh, parent);
+          resp.path = path;
+          created_path = path;
         }
       }
-      z = zxid_counter;
+      (void)child_watch_parent;
+    }
+    if (made_ephemeral) {
+      // registered outside the shard locks; if the session died in between,
+      // roll the node back (the sweeper can no longer see it)
+      bool ok = false;
+      SessionPtr s = c->session;
+      {
+        std::lock_guard<std::mutex> g(session_mu);
+        if (s && sessions.count(s->id)) {
+          std::lock_guard<std::mutex> eg(s->eph_mu);
+          s->ephemerals.insert(created_path);
+          ok = true;
+        }
+      }
+      if (!ok) {
+        delete_node(created_path);
+        err = kZSessionExpired;
+      }
     }
     if (err == kZOk)
-      send_reply(c, xid, z, kZOk, [&](JuteWriter& w) { resp.serialize(w); });
+      send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), kZOk,
+                 [&](JuteWriter& w) { resp.serialize(w); });
     else
-      send_reply(c, xid, z, err, nullptr);
+      send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), err, nullptr);
   }
 
   void handle_delete(Conn* c, int32_t xid, JuteReader& r) {
-    ops.del++;
+    ops.del.fetch_add(1, std::memory_order_relaxed);
     DeleteRequest req;
     req.deserialize(r);
     int32_t err = kZOk;
-    int64_t z;
     {
-      std::lock_guard<std::mutex> g(mu);
-      touch_session_locked(c->session_id);
-      auto it = nodes.find(req.path);
-      if (it == nodes.end() || req.path == "/") {
+      // peek under the target's shard lock; full delete re-locks both shards
+      Shard& sh = shard_of(req.path);
+      std::unique_lock<std::mutex> lk(sh.mu);
+      auto it = sh.nodes.find(req.path);
+      if (it == sh.nodes.end() || req.path == "/") {
         err = kZNoNode;
       } else if (!it->second.children.empty()) {
         err = kZNotEmpty;
       } else if (req.version != -1 && req.version != it->second.stat.version) {
         err = kZBadVersion;
-      } else {
-        delete_node_locked(req.path);
       }
-      z = zxid_counter;
     }
-    send_reply(c, xid, z, err, nullptr);
+    if (err == kZOk) {
+      if (!delete_node(req.path)) err = kZNoNode;  // raced with another delete
+    }
+    send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), err, nullptr);
   }
 
-  // mu held; fires watches, updates parent
-  void delete_node_locked(const std::string& path) {
-    auto it = nodes.find(path);
-    if (it == nodes.end()) return;
-    int64_t z = ++zxid_counter;
-    if (it->second.stat.ephemeral_owner != 0) {
-      auto eit = ephemerals.find(it->second.stat.ephemeral_owner);
-      if (eit != ephemerals.end()) eit->second.erase(path);
+  // Full node removal: locks child+parent shards, updates parent, fires
+  // watches, detaches from the owner session. Returns false if missing.
+  bool delete_node(const std::string& path) {
+    int64_t owner = 0;
+    {
+      std::string parent = parent_path(path);
+      TwoShardLock lk(*this, path, parent);
+      Shard& csh = shard_of(path);
+      auto it = csh.nodes.find(path);
+      if (it == csh.nodes.end()) return false;
+      if (!it->second.children.empty()) return false;  // re-check under lock
+      int64_t z = zxid_counter.fetch_add(1, std::memory_order_relaxed) + 1;
+      owner = it->second.stat.ephemeral_owner;
+      csh.nodes.erase(it);
+      Shard& psh = shard_of(parent);
+      auto pit = psh.nodes.find(parent);
+      if (pit != psh.nodes.end()) {
+        pit->second.children.erase(basename_of(path));
+        pit->second.stat.cversion++;
+        pit->second.stat.pzxid = z;
+        pit->second.stat.num_children = static_cast<int32_t>(pit->second.children.size());
+        fire_child_watches_locked(psh, parent);
+      }
+      fire_data_watches_locked(csh, path, kEventNodeDeleted);
     }
-    nodes.erase(it);
-    std::string parent = parent_path(path);
-    auto pit = nodes.find(parent);
-    if (pit != nodes.end()) {
-      pit->second.children.erase(basename_of(path));
-      pit->second.stat.cversion++;
-      pit->second.stat.pzxid = z;
-      pit->second.stat.num_children = static_cast<int32_t>(pit->second.children.size());
-      fire_child_watches_locked(parent);
+    if (owner != 0) {
+      SessionPtr s;
+      {
+        std::lock_guard<std::mutex> g(session_mu);
+        auto sit = sessions.find(owner);
+        if (sit != sessions.end()) s = sit->second;
+      }
+      if (s) {
+        std::lock_guard<std::mutex> eg(s->eph_mu);
+        s->ephemerals.erase(path);
+      }
     }
-    fire_data_watches_locked(path, kEventNodeDeleted);
+    return true;
   }
 
   void handle_exists(Conn* c, int32_t xid, JuteReader& r) {
-    ops.exists++;
+    ops.exists.fetch_add(1, std::memory_order_relaxed);
     ExistsRequest req;
     req.deserialize(r);
     int32_t err = kZOk;
-    int64_t z;
     ExistsResponse resp;
     {
-      std::lock_guard<std::mutex> g(mu);
-      touch_session_locked(c->session_id);
-      if (req.watch) data_watches[req.path].insert(c->session_id);
-      auto it = nodes.find(req.path);
-      if (it == nodes.end())
+      Shard& sh = shard_of(req.path);
+      std::lock_guard<std::mutex> lk(sh.mu);
+      if (req.watch) sh.data_watches[req.path].insert(sid_of(c));
+      auto it = sh.nodes.find(req.path);
+      if (it == sh.nodes.end())
         err = kZNoNode;
       else
         resp.stat = it->second.stat;
-      z = zxid_counter;
     }
     if (err == kZOk)
-      send_reply(c, xid, z, kZOk, [&](JuteWriter& w) { resp.serialize(w); });
+      send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), kZOk,
+                 [&](JuteWriter& w) { resp.serialize(w); });
     else
-      send_reply(c, xid, z, err, nullptr);
+      send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), err, nullptr);
   }
 
   void handle_get_data(Conn* c, int32_t xid, JuteReader& r) {
-    ops.get_data++;
+    ops.get_data.fetch_add(1, std::memory_order_relaxed);
     GetDataRequest req;
     req.deserialize(r);
     int32_t err = kZOk;
-    int64_t z;
     GetDataResponse resp;
     {
-      std::lock_guard<std::mutex> g(mu);
-      touch_session_locked(c->session_id);
-      auto it = nodes.find(req.path);
-      if (it == nodes.end()) {
+      Shard& sh = shard_of(req.path);
+      std::lock_guard<std::mutex> lk(sh.mu);
+      auto it = sh.nodes.find(req.path);
+      if (it == sh.nodes.end()) {
         err = kZNoNode;
       } else {
-        if (req.watch) data_watches[req.path].insert(c->session_id);
+        if (req.watch) sh.data_watches[req.path].insert(sid_of(c));
         resp.data = it->second.data;
         resp.stat = it->second.stat;
       }
-      z = zxid_counter;
     }
     if (err == kZOk)
-      send_reply(c, xid, z, kZOk, [&](JuteWriter& w) { resp.serialize(w); });
+      send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), kZOk,
+                 [&](JuteWriter& w) { resp.serialize(w); });
     else
-      send_reply(c, xid, z, err, nullptr);
+      send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), err, nullptr);
   }
 
   void handle_set_data(Conn* c, int32_t xid, JuteReader& r) {
-    ops.set_data++;
+    ops.set_data.fetch_add(1, std::memory_order_relaxed);
     SetDataRequest req;
     req.deserialize(r);
     int32_t err = kZOk;
-    int64_t z;
     SetDataResponse resp;
     {
-      std::lock_guard<std::mutex> g(mu);
-      touch_session_locked(c->session_id);
-      auto it = nodes.find(req.path);
-      if (it == nodes.end()) {
+      Shard& sh = shard_of(req.path);
+      std::lock_guard<std::mutex> lk(sh.mu);
+      auto it = sh.nodes.find(req.path);
+      if (it == sh.nodes.end()) {
         err = kZNoNode;
       } else if (req.version != -1 && req.version != it->second.stat.version) {
         err = kZBadVersion;
       } else {
-        int64_t zz = ++zxid_counter;
+        int64_t zz = zxid_counter.fetch_add(1, std::memory_order_relaxed) + 1;
         it->second.data = req.data;
         it->second.stat.mzxid = zz;
         it->second.stat.mtime = wall_ms();
         it->second.stat.version++;
         it->second.stat.data_length = static_cast<int32_t>(req.data.size());
-        fire_data_watches_locked(req.path, kEventNodeDataChanged);
+        fire_data_watches_locked(sh, req.path, kEventNodeDataChanged);
         resp.stat = it->second.stat;
       }
-      z = zxid_counter;
     }
     if (err == kZOk)
-      send_reply(c, xid, z, kZOk, [&](JuteWriter& w) { resp.serialize(w); });
+      send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), kZOk,
+                 [&](JuteWriter& w) { resp.serialize(w); });
     else
-      send_reply(c, xid, z, err, nullptr);
+      send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), err, nullptr);
   }
 
   void handle_get_children(Conn* c, int32_t xid, JuteReader& r) {
-    ops.get_children++;
+    ops.get_children.fetch_add(1, std::memory_order_relaxed);
     GetChildrenRequest req;
     req.deserialize(r);
     int32_t err = kZOk;
-    int64_t z;
     GetChildrenResponse resp;
     {
-      std::lock_guard<std::mutex> g(mu);
-      touch_session_locked(c->session_id);
-      auto it = nodes.find(req.path);
-      if (it == nodes.end()) {
+      Shard& sh = shard_of(req.path);
+      std::lock_guard<std::mutex> lk(sh.mu);
+      auto it = sh.nodes.find(req.path);
+      if (it == sh.nodes.end()) {
         err = kZNoNode;
       } else {
-        if (req.watch) child_watches[req.path].insert(c->session_id);
+        if (req.watch) sh.child_watches[req.path].insert(sid_of(c));
         resp.children.assign(it->second.children.begin(), it->second.children.end());
       }
-      z = zxid_counter;
     }
     if (err == kZOk)
-      send_reply(c, xid, z, kZOk, [&](JuteWriter& w) { resp.serialize(w); });
+      send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), kZOk,
+                 [&](JuteWriter& w) { resp.serialize(w); });
     else
-      send_reply(c, xid, z, err, nullptr);
+      send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), err, nullptr);
   }
 
   void handle_close_session(Conn* c, int32_t xid) {
-    ops.close_session++;
-    int64_t sid = c->session_id;
-    int64_t z;
-    {
-      std::lock_guard<std::mutex> g(mu);
-      z = zxid_counter;
-    }
+    ops.close_session.fetch_add(1, std::memory_order_relaxed);
+    int64_t sid = sid_of(c);
     {
       std::lock_guard<std::mutex> og(c->out_mu);
       c->closing = true;  // close after the reply drains
     }
-    send_reply(c, xid, z, kZOk, nullptr);
-    if (sid != 0) {
-      std::lock_guard<std::mutex> g(mu);
-      kill_session_locked(sid, /*close_conn=*/false);
-    }
+    send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), kZOk, nullptr);
+    if (sid != 0) kill_session(sid, /*close_conn_too=*/false);
   }
 
   // ---------------- session lifecycle ----------------
 
-  // mu held
-  void touch_session_locked(int64_t sid) {
-    auto it = sessions.find(sid);
-    if (it != sessions.end()) it->second.last_touch = now_ms();
-  }
-
-  // mu held. Expire/close a session: remove ephemerals (firing watches),
-  // tombstone it, optionally retire its connection.
-  void kill_session_locked(int64_t sid, bool close_conn_too) {
-    auto it = sessions.find(sid);
-    if (it == sessions.end()) return;
-    uint64_t cid = it->second.conn_id;
-    auto eit = ephemerals.find(sid);
-    if (eit != ephemerals.end()) {
-      std::vector<std::string> paths(eit->second.begin(), eit->second.end());
-      for (const auto& p : paths) delete_node_locked(p);
-      ephemerals.erase(sid);
+  // Expire/close a session: remove ephemerals (firing watches), tombstone it,
+  // optionally retire its connection. Lock discipline: session_mu alone, then
+  // eph_mu alone, then per-path shard locks via delete_node.
+  void kill_session(int64_t sid, bool close_conn_too) {
+    SessionPtr s;
+    {
+      std::lock_guard<std::mutex> g(session_mu);
+      auto it = sessions.find(sid);
+      if (it == sessions.end()) return;
+      s = it->second;
+      sessions.erase(it);
+      dead_sessions.insert(sid);
     }
-    for (auto& kv : data_watches) kv.second.erase(sid);
-    for (auto& kv : child_watches) kv.second.erase(sid);
-    sessions.erase(sid);
-    dead_sessions.insert(sid);
+    std::set<std::string> eph;
+    {
+      std::lock_guard<std::mutex> eg(s->eph_mu);
+      eph.swap(s->ephemerals);
+    }
+    for (const auto& p : eph) delete_node(p);
+    // stale watch registrations for this session are skipped at fire time
+    uint64_t cid = s->conn_id.load();
     if (close_conn_too && cid != 0) {
-      auto cit = conns.find(cid);
-      if (cit != conns.end()) {
-        cit->second->dead.store(true);
-        post_close_locked(cit->second);
+      ConnPtr cp = lookup(cid);
+      if (cp) {
+        cp->dead.store(true);
+        post_close(cp);
       }
     }
   }
 
   void schedule_sweep() {
     loops[0]->schedule(cfg.tick_ms, [this] {
+      std::vector<int64_t> expired;
       {
-        std::lock_guard<std::mutex> g(mu);
+        std::lock_guard<std::mutex> g(session_mu);
         int64_t now = now_ms();
-        std::vector<int64_t> expired;
         for (const auto& kv : sessions)
-          if (now - kv.second.last_touch > kv.second.timeout_ms) expired.push_back(kv.first);
-        for (int64_t sid : expired) {
-          log.info("session expired", {{"session", Json(sid)}});
-          kill_session_locked(sid, /*close_conn=*/true);
-        }
+          if (now - kv.second->last_touch.load(std::memory_order_relaxed) > kv.second->timeout_ms)
+            expired.push_back(kv.first);
+      }
+      for (int64_t sid : expired) {
+        log.info("session expired", {{"session", Json(sid)}});
+        kill_session(sid, /*close_conn_too=*/true);
       }
       schedule_sweep();
     });
   }
 
-  // ---------------- watches ----------------
+  // ---------------- watches (shard lock held) ----------------
 
-  // mu held for all three
-  void fire_data_watches_locked(const std::string& path, int32_t event_type) {
-    auto it = data_watches.find(path);
-    if (it == data_watches.end()) return;
+  void fire_data_watches_locked(Shard& sh, const std::string& path, int32_t event_type) {
+    auto it = sh.data_watches.find(path);
+    if (it == sh.data_watches.end()) return;
     std::set<int64_t> watchers = std::move(it->second);
-    data_watches.erase(it);
-    deliver_watch_locked(watchers, path, event_type);
+    sh.data_watches.erase(it);
+    deliver_watch(watchers, path, event_type);
   }
 
-  void fire_child_watches_locked(const std::string& path) {
-    auto it = child_watches.find(path);
-    if (it == child_watches.end()) return;
+  void fire_child_watches_locked(Shard& sh, const std::string& path) {
+    auto it = sh.child_watches.find(path);
+    if (it == sh.child_watches.end()) return;
     std::set<int64_t> watchers = std::move(it->second);
-    child_watches.erase(it);
-    deliver_watch_locked(watchers, path, kEventNodeChildrenChanged);
+    sh.child_watches.erase(it);
+    deliver_watch(watchers, path, kEventNodeChildrenChanged);
   }
 
-  void deliver_watch_locked(const std::set<int64_t>& watchers, const std::string& path, int32_t event_type) {
+  void deliver_watch(const std::set<int64_t>& watchers, const std::string& path, int32_t event_type) {
     for (int64_t sid : watchers) {
-      auto sit = sessions.find(sid);
-      if (sit == sessions.end() || sit->second.conn_id == 0) continue;
-      auto cit = conns.find(sit->second.conn_id);
-      if (cit == conns.end()) continue;
+      uint64_t cid = 0;
+      {
+        std::lock_guard<std::mutex> g(session_mu);
+        auto sit = sessions.find(sid);
+        if (sit == sessions.end()) continue;  // stale registration
+        cid = sit->second->conn_id.load();
+      }
+      if (cid == 0) continue;
+      ConnPtr cp = lookup(cid);
+      if (!cp) continue;
       std::string pkt;
       begin_packet(&pkt);
       JuteWriter w(&pkt);
@@ -788,7 +865,7 @@ struct Ensemble::Impl {
       ev.path = path;
       ev.serialize(w);
       frame_packet(&pkt);
-      send_packet(cit->second.get(), std::move(pkt));
+      send_packet(cp.get(), std::move(pkt));
     }
   }
 
@@ -851,8 +928,8 @@ struct Ensemble::Impl {
     deliver(c, std::move(pkt));
   }
 
-  // Any thread. Direct write when the buffer is clear, else buffer + arm
-  // EPOLLOUT on the owner loop.
+  // Cross-thread delivery: direct write when the buffer is clear, else
+  // buffer + arm EPOLLOUT on the owner loop.
   void deliver(Conn* c, std::string pkt) {
     std::lock_guard<std::mutex> og(c->out_mu);
     if (c->dead.load()) return;
@@ -890,8 +967,7 @@ struct Ensemble::Impl {
     }
   }
 
-  // out_mu held (or unambiguous owner). Marks the conn for closure; the owner
-  // loop performs the actual close.
+  // Marks the conn for closure; the owner loop performs the actual close.
   void mark_dead(Conn* c) {
     if (c->dead.exchange(true)) return;
     uint64_t cid = c->id;
@@ -922,9 +998,13 @@ struct Ensemble::Impl {
       if (!close_now) {
         c->outbuf.erase(0, off);
         if (c->outbuf.empty()) {
-          c->epollout_armed = false;
-          loops[c->server_idx]->mod_fd(c->fd, EPOLLIN);
+          if (c->epollout_armed) {
+            c->epollout_armed = false;
+            loops[c->server_idx]->mod_fd(c->fd, EPOLLIN);
+          }
           if (c->closing) close_now = true;
+        } else {
+          arm_epollout(c);
         }
       }
     }
@@ -934,20 +1014,28 @@ struct Ensemble::Impl {
   // MUST run on the owner loop thread.
   void close_conn(Conn* c) {
     c->dead.store(true);
-    std::lock_guard<std::mutex> g(mu);
-    auto it = conns.find(c->id);
-    if (it == conns.end()) return;  // already closed
-    if (c->session_id != 0) {
-      auto sit = sessions.find(c->session_id);
-      if (sit != sessions.end() && sit->second.conn_id == c->id) sit->second.conn_id = 0;
+    {
+      std::lock_guard<std::mutex> g(conns_mu);
+      auto it = conns.find(c->id);
+      if (it == conns.end()) return;  // already closed
+      conns.erase(it);
     }
-    loops[c->server_idx]->del_fd(c->fd);
-    ::close(c->fd);
-    conns.erase(it);
+    if (c->session) {
+      uint64_t expected = c->id;
+      c->session->conn_id.compare_exchange_strong(expected, 0);
+    }
+    {
+      // out_mu serializes against a cross-thread deliver() mid-write: the fd
+      // must not be closed (and possibly reused) under a concurrent write
+      std::lock_guard<std::mutex> og(c->out_mu);
+      loops[c->server_idx]->del_fd(c->fd);
+      ::close(c->fd);
+      c->fd = -1;
+    }
   }
 
-  // mu held: queue a close on the conn's owner loop
-  void post_close_locked(const ConnPtr& c) {
+  // queue a close on the conn's owner loop
+  void post_close(const ConnPtr& c) {
     uint64_t cid = c->id;
     size_t srv = c->server_idx;
     loops[srv]->post([this, cid] {
@@ -962,21 +1050,28 @@ struct Ensemble::Impl {
     if (idx >= loops.size()) return;
     std::promise<void> done;
     loops[idx]->post([this, idx, &done] {
-      std::vector<ConnPtr> victims;
+      bool was_up = false;
       {
-        std::lock_guard<std::mutex> g(mu);
+        std::lock_guard<std::mutex> g(admin_mu);
         if (idx < servers.size() && servers[idx].up) {
+          was_up = true;
           Server& s = servers[idx];
           loops[idx]->del_fd(s.listen_fd);
           ::close(s.listen_fd);
           s.listen_fd = -1;
           s.up = false;
+        }
+      }
+      if (was_up) {
+        std::vector<ConnPtr> victims;
+        {
+          std::lock_guard<std::mutex> g(conns_mu);
           for (auto& kv : conns)
             if (kv.second->server_idx == idx) victims.push_back(kv.second);
         }
+        for (auto& c : victims) close_conn(c.get());
+        log.info("server killed", {{"server", Json(static_cast<int64_t>(idx))}});
       }
-      for (auto& c : victims) close_conn(c.get());
-      log.info("server killed", {{"server", Json(static_cast<int64_t>(idx))}});
       done.set_value();
     });
     done.get_future().wait();
@@ -987,7 +1082,7 @@ struct Ensemble::Impl {
     std::promise<void> done;
     loops[idx]->post([this, idx, &done] {
       {
-        std::lock_guard<std::mutex> g(mu);
+        std::lock_guard<std::mutex> g(admin_mu);
         if (idx < servers.size() && !servers[idx].up) {
           open_listener(idx, servers[idx].port);
           log.info("server restarted", {{"server", Json(static_cast<int64_t>(idx))}});
@@ -998,7 +1093,7 @@ struct Ensemble::Impl {
     done.get_future().wait();
   }
 
-  // mu held
+  // admin_mu held
   void maybe_elect_leader_locked() {
     if (leader_idx < servers.size() && servers[leader_idx].up) return;
     for (size_t i = 0; i < servers.size(); i++) {
@@ -1013,19 +1108,14 @@ struct Ensemble::Impl {
   size_t kill_leader() {
     size_t victim;
     {
-      std::lock_guard<std::mutex> g(mu);
+      std::lock_guard<std::mutex> g(admin_mu);
       victim = leader_idx;
     }
     kill_server(victim);
-    std::lock_guard<std::mutex> g(mu);
+    std::lock_guard<std::mutex> g(admin_mu);
     if (cfg.election_ms > 0) election_until = now_ms() + cfg.election_ms;
     maybe_elect_leader_locked();
     return victim;
-  }
-
-  void expire_session(int64_t session_id) {
-    std::lock_guard<std::mutex> g(mu);
-    kill_session_locked(session_id, /*close_conn=*/true);
   }
 };
 
@@ -1044,7 +1134,7 @@ void Ensemble::start() { impl_->start(); }
 void Ensemble::stop() { impl_->stop(); }
 
 std::vector<int> Ensemble::ports() const {
-  std::lock_guard<std::mutex> g(impl_->mu);
+  std::lock_guard<std::mutex> g(impl_->admin_mu);
   std::vector<int> out;
   for (const auto& s : impl_->servers) out.push_back(s.port);
   return out;
@@ -1063,26 +1153,27 @@ void Ensemble::kill_server(size_t idx) { impl_->kill_server(idx); }
 void Ensemble::restart_server(size_t idx) { impl_->restart_server(idx); }
 
 bool Ensemble::server_up(size_t idx) const {
-  std::lock_guard<std::mutex> g(impl_->mu);
+  std::lock_guard<std::mutex> g(impl_->admin_mu);
   return idx < impl_->servers.size() && impl_->servers[idx].up;
 }
 
 size_t Ensemble::leader() const {
-  std::lock_guard<std::mutex> g(impl_->mu);
+  std::lock_guard<std::mutex> g(impl_->admin_mu);
   return impl_->leader_idx;
 }
 
 size_t Ensemble::kill_leader() { return impl_->kill_leader(); }
 
-void Ensemble::expire_session(int64_t session_id) { impl_->expire_session(session_id); }
+void Ensemble::expire_session(int64_t session_id) { impl_->kill_session(session_id, /*close_conn_too=*/true); }
 
 void Ensemble::set_latency_ms(int ms) { impl_->latency_ms.store(ms); }
 
 NodeInfo Ensemble::get(const std::string& path) const {
-  std::lock_guard<std::mutex> g(impl_->mu);
+  auto& sh = impl_->shard_of(path);
+  std::lock_guard<std::mutex> g(sh.mu);
   NodeInfo info;
-  auto it = impl_->nodes.find(path);
-  if (it != impl_->nodes.end()) {
+  auto it = sh.nodes.find(path);
+  if (it != sh.nodes.end()) {
     info.exists = true;
     info.data = it->second.data;
     info.stat = it->second.stat;
@@ -1091,35 +1182,44 @@ NodeInfo Ensemble::get(const std::string& path) const {
 }
 
 std::vector<std::string> Ensemble::children(const std::string& path) const {
-  std::lock_guard<std::mutex> g(impl_->mu);
-  auto it = impl_->nodes.find(path);
-  if (it == impl_->nodes.end()) return {};
+  auto& sh = impl_->shard_of(path);
+  std::lock_guard<std::mutex> g(sh.mu);
+  auto it = sh.nodes.find(path);
+  if (it == sh.nodes.end()) return {};
   return std::vector<std::string>(it->second.children.begin(), it->second.children.end());
 }
 
 size_t Ensemble::node_count() const {
-  std::lock_guard<std::mutex> g(impl_->mu);
-  return impl_->nodes.size() - 1;  // exclude root
+  size_t n = 0;
+  for (auto& sh : impl_->shards) {
+    std::lock_guard<std::mutex> g(sh.mu);
+    n += sh.nodes.size();
+  }
+  return n - 1;  // exclude root
 }
 
 size_t Ensemble::ephemeral_count() const {
-  std::lock_guard<std::mutex> g(impl_->mu);
+  std::vector<Impl::SessionPtr> snap;
+  {
+    std::lock_guard<std::mutex> g(impl_->session_mu);
+    for (const auto& kv : impl_->sessions) snap.push_back(kv.second);
+  }
   size_t n = 0;
-  for (const auto& kv : impl_->ephemerals) n += kv.second.size();
+  for (const auto& s : snap) {
+    std::lock_guard<std::mutex> eg(s->eph_mu);
+    n += s->ephemerals.size();
+  }
   return n;
 }
 
 std::vector<int64_t> Ensemble::session_ids() const {
-  std::lock_guard<std::mutex> g(impl_->mu);
+  std::lock_guard<std::mutex> g(impl_->session_mu);
   std::vector<int64_t> out;
   for (const auto& kv : impl_->sessions) out.push_back(kv.first);
   return out;
 }
 
-int64_t Ensemble::zxid() const {
-  std::lock_guard<std::mutex> g(impl_->mu);
-  return impl_->zxid_counter;
-}
+int64_t Ensemble::zxid() const { return impl_->zxid_counter.load(); }
 
 std::map<std::string, uint64_t> Ensemble::counters() const {
   std::map<std::string, uint64_t> out;
@@ -1134,7 +1234,6 @@ std::map<std::string, uint64_t> Ensemble::counters() const {
   out["ping"] = o.ping.load();
   out["closeSession"] = o.close_session.load();
   out["unknown"] = o.unknown.load();
-  // drop zero counters so tests can assert presence meaningfully
   for (auto it = out.begin(); it != out.end();) {
     if (it->second == 0)
       it = out.erase(it);
