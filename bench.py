@@ -84,7 +84,9 @@ def main():
 
     # ---- ensemble: hosted by rank 0, shared over localhost TCP ----
     if args.servers <= 0:
-        args.servers = 1 if world == 1 else 3
+        # production ZK quorum sizes: 1 standalone / 3 / 5 (BASELINE configs
+        # 1 vs 2/3; 5-node quorums are standard at 8-client scale)
+        args.servers = 1 if world == 1 else (3 if world <= 4 else 5)
     ensemble = None
     if rank == 0:
         ensemble = ra.Ensemble(servers=args.servers, tick_ms=100, max_session_timeout_ms=60000)
@@ -102,8 +104,11 @@ def main():
     for hp in connect.split(","):
         host, port = hp.rsplit(":", 1)
         servers.append((host, int(port)))
+    # deterministic load balance: rank r starts at server r % len(servers)
+    rot = rank % len(servers)
+    servers = servers[rot:] + servers[:rot]
 
-    client = ra.ZkClient(servers=servers, session_timeout_ms=40000)
+    client = ra.ZkClient(servers=servers, session_timeout_ms=40000, randomize_start=False)
     client.start()
     if not client.wait_connected(30000):
         raise RuntimeError("rank %d: could not connect to ensemble" % rank)

@@ -84,7 +84,8 @@ py::dict health_record_to_dict(const HealthRecord& rec) {
 zk::ZkClientConfig client_config_from_args(const std::vector<std::pair<std::string, int>>& servers,
                                            int session_timeout_ms, int connect_timeout_ms,
                                            int64_t connect_initial_delay_ms, int64_t connect_max_delay_ms,
-                                           int64_t connect_max_attempts, const std::string& log_level) {
+                                           int64_t connect_max_attempts, bool randomize_start,
+                                           const std::string& log_level) {
   zk::ZkClientConfig cfg;
   for (const auto& s : servers) cfg.servers.push_back({s.first, s.second});
   cfg.session_timeout_ms = session_timeout_ms;
@@ -92,6 +93,7 @@ zk::ZkClientConfig client_config_from_args(const std::vector<std::pair<std::stri
   cfg.connect_initial_delay_ms = connect_initial_delay_ms;
   cfg.connect_max_delay_ms = connect_max_delay_ms;
   cfg.connect_max_attempts = connect_max_attempts;
+  cfg.randomize_start = randomize_start;
   cfg.log_level = level_from(log_level);
   return cfg;
 }
@@ -178,15 +180,16 @@ PYBIND11_MODULE(_core, m) {
   py::class_<zk::ZkClient>(m, "ZkClient")
       .def(py::init([](const std::vector<std::pair<std::string, int>>& servers, int session_timeout_ms,
                        int connect_timeout_ms, int64_t connect_initial_delay_ms, int64_t connect_max_delay_ms,
-                       int64_t connect_max_attempts, const std::string& log_level) {
+                       int64_t connect_max_attempts, bool randomize_start, const std::string& log_level) {
              return std::make_unique<zk::ZkClient>(
                  client_config_from_args(servers, session_timeout_ms, connect_timeout_ms, connect_initial_delay_ms,
-                                         connect_max_delay_ms, connect_max_attempts, log_level),
+                                         connect_max_delay_ms, connect_max_attempts, randomize_start, log_level),
                  make_logger("zkclient", log_level));
            }),
            py::arg("servers"), py::arg("session_timeout_ms") = 30000, py::arg("connect_timeout_ms") = 4000,
            py::arg("connect_initial_delay_ms") = 1000, py::arg("connect_max_delay_ms") = 90000,
-           py::arg("connect_max_attempts") = -1, py::arg("log_level") = "warn")
+           py::arg("connect_max_attempts") = -1, py::arg("randomize_start") = true,
+           py::arg("log_level") = "warn")
       .def("start", &zk::ZkClient::start, py::call_guard<py::gil_scoped_release>())
       .def("wait_connected", &zk::ZkClient::wait_connected, py::arg("timeout_ms") = -1,
            py::call_guard<py::gil_scoped_release>())

@@ -93,6 +93,7 @@ struct Ensemble::Impl {
     int timeout_ms = 30000;
     std::atomic<int64_t> last_touch{0};  // monotonic ms, lock-free touch
     std::atomic<uint64_t> conn_id{0};    // 0 = detached
+    std::atomic<bool> alive{true};       // false once kill_session starts
     std::mutex eph_mu;
     std::set<std::string> ephemerals;
   };
@@ -568,20 +569,26 @@ struct Ensemble::Impl {
       (void)child_watch_parent;
     }
     if (made_ephemeral) {
-      // registered outside the shard locks; if the session died in between,
-      // roll the node back (the sweeper can no longer see it)
-      bool ok = false;
+      // Registered outside the shard locks — NO global lock on this hot path.
+      // kill_session flips `alive` BEFORE draining the set, so: insert seen
+      // by the drain ⇒ cleaned there; insert after the drain ⇒ alive==false
+      // here ⇒ roll the node back ourselves.
       SessionPtr s = c->session;
-      {
-        std::lock_guard<std::mutex> g(session_mu);
-        if (s && sessions.count(s->id)) {
+      bool ok = false;
+      if (s) {
+        {
           std::lock_guard<std::mutex> eg(s->eph_mu);
           s->ephemerals.insert(created_path);
-          ok = true;
+        }
+        ok = s->alive.load(std::memory_order_acquire);
+        if (!ok) {
+          std::lock_guard<std::mutex> eg(s->eph_mu);
+          ok = s->ephemerals.count(created_path) == 0;  // drain took it: fine
+          if (!ok) s->ephemerals.erase(created_path);
         }
       }
       if (!ok) {
-        delete_node(created_path);
+        delete_node(created_path, s);
         err = kZSessionExpired;
       }
     }
@@ -611,14 +618,17 @@ struct Ensemble::Impl {
       }
     }
     if (err == kZOk) {
-      if (!delete_node(req.path)) err = kZNoNode;  // raced with another delete
+      // common case: a session unlinking its own ephemerals (cleanup step)
+      if (!delete_node(req.path, c->session)) err = kZNoNode;  // raced
     }
     send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), err, nullptr);
   }
 
   // Full node removal: locks child+parent shards, updates parent, fires
   // watches, detaches from the owner session. Returns false if missing.
-  bool delete_node(const std::string& path) {
+  // owner_hint avoids the session_mu map lookup on the hot path (a session
+  // deleting its own ephemerals — the register-pipeline cleanup case).
+  bool delete_node(const std::string& path, const SessionPtr& owner_hint = nullptr) {
     int64_t owner = 0;
     {
       std::string parent = parent_path(path);
@@ -643,7 +653,9 @@ struct Ensemble::Impl {
     }
     if (owner != 0) {
       SessionPtr s;
-      {
+      if (owner_hint && owner_hint->id == owner) {
+        s = owner_hint;
+      } else {
         std::lock_guard<std::mutex> g(session_mu);
         auto sit = sessions.find(owner);
         if (sit != sessions.end()) s = sit->second;
@@ -786,12 +798,13 @@ struct Ensemble::Impl {
       sessions.erase(it);
       dead_sessions.insert(sid);
     }
+    s->alive.store(false, std::memory_order_release);
     std::set<std::string> eph;
     {
       std::lock_guard<std::mutex> eg(s->eph_mu);
       eph.swap(s->ephemerals);
     }
-    for (const auto& p : eph) delete_node(p);
+    for (const auto& p : eph) delete_node(p, s);
     // stale watch registrations for this session are skipped at fire time
     uint64_t cid = s->conn_id.load();
     if (close_conn_too && cid != 0) {

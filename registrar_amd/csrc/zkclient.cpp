@@ -99,9 +99,10 @@ struct ZkClient::Impl {
     log.set_level(cfg.log_level);
     // spread clients across the ensemble: random starting server (real ZK
     // clients shuffle the connect string for the same reason)
-    server_idx = static_cast<size_t>((static_cast<uint64_t>(getpid()) * 0x9E3779B97F4A7C15ull +
-                                      reinterpret_cast<uintptr_t>(this)) >>
-                                     17);
+    if (cfg.randomize_start)
+      server_idx = static_cast<size_t>((static_cast<uint64_t>(getpid()) * 0x9E3779B97F4A7C15ull +
+                                        reinterpret_cast<uintptr_t>(this)) >>
+                                       17);
     connect_backoff.initial_ms = cfg.connect_initial_delay_ms;
     connect_backoff.max_ms = cfg.connect_max_delay_ms;
     connect_backoff.max_attempts = cfg.connect_max_attempts;

@@ -65,6 +65,9 @@ struct ZkClientConfig {
   // session-preserving reconnect retry (fast: the session clock is ticking)
   int64_t reconnect_initial_delay_ms = 10;
   int64_t reconnect_max_delay_ms = 1000;
+  // start at a random server (ZK-client convention; set false when the
+  // caller pre-ordered the list for deterministic load balance)
+  bool randomize_start = true;
   LogLevel log_level = LogLevel::Warn;
 };
 
