@@ -102,7 +102,8 @@ struct Ensemble::Impl {
   struct Conn {
     uint64_t id = 0;
     int fd = -1;
-    size_t server_idx = 0;
+    size_t server_idx = 0;  // which server (port) accepted it — kill semantics
+    size_t io_idx = 0;      // which IO loop owns its events
     // owner-loop-thread only:
     std::string inbuf;
     size_t inpos = 0;
@@ -131,8 +132,9 @@ struct Ensemble::Impl {
 
   EnsembleConfig cfg;
   Logger log;
-  std::vector<std::unique_ptr<EventLoop>> loops;  // one per server slot
+  std::vector<std::unique_ptr<EventLoop>> loops;  // IO pool (>= #servers)
   std::vector<std::thread> threads;
+  std::atomic<size_t> next_io{0};
   std::atomic<bool> started{false};
   std::atomic<int> latency_ms{0};
   std::atomic<int64_t> zxid_counter{0};
@@ -186,14 +188,20 @@ struct Ensemble::Impl {
   void start() {
     if (started.exchange(true)) return;
     size_t n = cfg.ports.size();
+    size_t io = static_cast<size_t>(cfg.io_threads);
+    if (io == 0) {
+      unsigned hw = std::thread::hardware_concurrency();
+      io = std::max(n, std::min<size_t>(8, hw ? hw / 4 : 2));
+      if (io == 0) io = 1;
+    }
     loops.clear();
-    for (size_t i = 0; i < n; i++) loops.push_back(std::make_unique<EventLoop>());
+    for (size_t i = 0; i < io; i++) loops.push_back(std::make_unique<EventLoop>());
     {
       std::lock_guard<std::mutex> g(admin_mu);
       servers.resize(n);
       for (size_t i = 0; i < n; i++) open_listener(i, cfg.ports[i]);
     }
-    for (size_t i = 0; i < n; i++) {
+    for (size_t i = 0; i < io; i++) {
       threads.emplace_back([this, i] {
         if (i == 0) schedule_sweep();
         loops[i]->run();
@@ -210,16 +218,18 @@ struct Ensemble::Impl {
         {
           std::lock_guard<std::mutex> g(conns_mu);
           for (auto& kv : conns)
-            if (kv.second->server_idx == i) victims.push_back(kv.second);
+            if (kv.second->io_idx == i) victims.push_back(kv.second);
         }
         for (auto& c : victims) close_conn(c.get());
         {
           std::lock_guard<std::mutex> g(admin_mu);
-          if (i < servers.size() && servers[i].listen_fd >= 0) {
-            loops[i]->del_fd(servers[i].listen_fd);
-            ::close(servers[i].listen_fd);
-            servers[i].listen_fd = -1;
-            servers[i].up = false;
+          for (size_t srv = 0; srv < servers.size(); srv++) {
+            if (srv % loops.size() == i && servers[srv].listen_fd >= 0) {
+              loops[i]->del_fd(servers[srv].listen_fd);
+              ::close(servers[srv].listen_fd);
+              servers[srv].listen_fd = -1;
+              servers[srv].up = false;
+            }
           }
         }
         done[i].set_value();
@@ -258,13 +268,14 @@ struct Ensemble::Impl {
     servers[idx].port = ntohs(addr.sin_port);
     servers[idx].up = true;
     size_t srv = idx;
-    auto install = [this, fd, srv] {
-      loops[srv]->add_fd(fd, EPOLLIN, [this, fd, srv](uint32_t) { on_accept(fd, srv); });
+    size_t lidx = idx % loops.size();
+    auto install = [this, fd, srv, lidx] {
+      loops[lidx]->add_fd(fd, EPOLLIN, [this, fd, srv](uint32_t) { on_accept(fd, srv); });
     };
-    if (loops[srv]->on_loop_thread()) {
+    if (loops[lidx]->on_loop_thread()) {
       install();
-    } else if (started.load() && loops[srv]->running()) {
-      loops[srv]->post(install);
+    } else if (started.load() && loops[lidx]->running()) {
+      loops[lidx]->post(install);
     } else {
       install();  // loop thread not started yet: direct registration is safe
     }
@@ -290,12 +301,14 @@ struct Ensemble::Impl {
       conn->id = next_conn_id.fetch_add(1);
       conn->fd = fd;
       conn->server_idx = server_idx;
+      conn->io_idx = next_io.fetch_add(1) % loops.size();
       uint64_t cid = conn->id;
+      size_t io_idx = conn->io_idx;
       {
         std::lock_guard<std::mutex> g(conns_mu);
         conns[cid] = conn;
       }
-      loops[server_idx]->add_fd(fd, EPOLLIN, [this, cid](uint32_t ev) { on_conn_event(cid, ev); });
+      loops[io_idx]->add_fd(fd, EPOLLIN, [this, cid](uint32_t ev) { on_conn_event(cid, ev); });
     }
   }
 
@@ -911,13 +924,13 @@ struct Ensemble::Impl {
     int lat = latency_ms.load();
     if (lat > 0) {
       uint64_t cid = c->id;
-      loops[c->server_idx]->schedule_from_any(lat, [this, cid, pkt = std::move(pkt)]() mutable {
+      loops[c->io_idx]->schedule_from_any(lat, [this, cid, pkt = std::move(pkt)]() mutable {
         ConnPtr cp = lookup(cid);
         if (cp) deliver(cp.get(), std::move(pkt));
       });
       return;
     }
-    if (loops[c->server_idx]->on_loop_thread()) {
+    if (loops[c->io_idx]->on_loop_thread()) {
       // reply path: queue only; flushed once per event drain (on_conn_event)
       // with a same-iteration 0 ms timer as the safety net for sends outside
       // the event path (sweep-timer watch deliveries)
@@ -929,7 +942,7 @@ struct Ensemble::Impl {
       if (!c->flush_scheduled) {
         c->flush_scheduled = true;
         uint64_t cid = c->id;
-        loops[c->server_idx]->schedule(0, [this, cid] {
+        loops[c->io_idx]->schedule(0, [this, cid] {
           ConnPtr cp = lookup(cid);
           if (!cp) return;
           cp->flush_scheduled = false;
@@ -976,7 +989,7 @@ struct Ensemble::Impl {
   void arm_epollout(Conn* c) {
     if (!c->epollout_armed) {
       c->epollout_armed = true;
-      loops[c->server_idx]->mod_fd(c->fd, EPOLLIN | EPOLLOUT);
+      loops[c->io_idx]->mod_fd(c->fd, EPOLLIN | EPOLLOUT);
     }
   }
 
@@ -984,7 +997,7 @@ struct Ensemble::Impl {
   void mark_dead(Conn* c) {
     if (c->dead.exchange(true)) return;
     uint64_t cid = c->id;
-    loops[c->server_idx]->post([this, cid] {
+    loops[c->io_idx]->post([this, cid] {
       ConnPtr cp = lookup(cid);
       if (cp) close_conn(cp.get());
     });
@@ -1013,7 +1026,7 @@ struct Ensemble::Impl {
         if (c->outbuf.empty()) {
           if (c->epollout_armed) {
             c->epollout_armed = false;
-            loops[c->server_idx]->mod_fd(c->fd, EPOLLIN);
+            loops[c->io_idx]->mod_fd(c->fd, EPOLLIN);
           }
           if (c->closing) close_now = true;
         } else {
@@ -1041,7 +1054,7 @@ struct Ensemble::Impl {
       // out_mu serializes against a cross-thread deliver() mid-write: the fd
       // must not be closed (and possibly reused) under a concurrent write
       std::lock_guard<std::mutex> og(c->out_mu);
-      loops[c->server_idx]->del_fd(c->fd);
+      loops[c->io_idx]->del_fd(c->fd);
       ::close(c->fd);
       c->fd = -1;
     }
@@ -1050,7 +1063,7 @@ struct Ensemble::Impl {
   // queue a close on the conn's owner loop
   void post_close(const ConnPtr& c) {
     uint64_t cid = c->id;
-    size_t srv = c->server_idx;
+    size_t srv = c->io_idx;
     loops[srv]->post([this, cid] {
       ConnPtr cp = lookup(cid);
       if (cp) close_conn(cp.get());
@@ -1060,40 +1073,55 @@ struct Ensemble::Impl {
   // ---------------- control (any thread) ----------------
 
   void kill_server(size_t idx) {
-    if (idx >= loops.size()) return;
+    {
+      std::lock_guard<std::mutex> g(admin_mu);
+      if (idx >= servers.size()) return;
+    }
+    size_t lidx = idx % loops.size();
     std::promise<void> done;
-    loops[idx]->post([this, idx, &done] {
+    loops[lidx]->post([this, idx, lidx, &done] {
       bool was_up = false;
       {
         std::lock_guard<std::mutex> g(admin_mu);
         if (idx < servers.size() && servers[idx].up) {
           was_up = true;
           Server& s = servers[idx];
-          loops[idx]->del_fd(s.listen_fd);
+          loops[lidx]->del_fd(s.listen_fd);
           ::close(s.listen_fd);
           s.listen_fd = -1;
           s.up = false;
         }
       }
-      if (was_up) {
-        std::vector<ConnPtr> victims;
-        {
-          std::lock_guard<std::mutex> g(conns_mu);
-          for (auto& kv : conns)
-            if (kv.second->server_idx == idx) victims.push_back(kv.second);
-        }
-        for (auto& c : victims) close_conn(c.get());
-        log.info("server killed", {{"server", Json(static_cast<int64_t>(idx))}});
-      }
       done.set_value();
     });
     done.get_future().wait();
+    // retire this server's connections on their own IO loops
+    std::vector<ConnPtr> victims;
+    {
+      std::lock_guard<std::mutex> g(conns_mu);
+      for (auto& kv : conns)
+        if (kv.second->server_idx == idx) victims.push_back(kv.second);
+    }
+    std::vector<std::promise<void>> vdone(victims.size());
+    for (size_t i = 0; i < victims.size(); i++) {
+      ConnPtr c = victims[i];
+      c->dead.store(true);
+      loops[c->io_idx]->post([this, c, &vdone, i] {
+        close_conn(c.get());
+        vdone[i].set_value();
+      });
+    }
+    for (auto& d : vdone) d.get_future().wait();
+    log.info("server killed", {{"server", Json(static_cast<int64_t>(idx))}});
   }
 
   void restart_server(size_t idx) {
-    if (idx >= loops.size()) return;
+    {
+      std::lock_guard<std::mutex> g(admin_mu);
+      if (idx >= servers.size()) return;
+    }
     std::promise<void> done;
-    loops[idx]->post([this, idx, &done] {
+    loops[idx % loops.size()]->post([this, idx, &done] {
       {
         std::lock_guard<std::mutex> g(admin_mu);
         if (idx < servers.size() && !servers[idx].up) {

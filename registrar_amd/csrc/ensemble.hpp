@@ -45,6 +45,10 @@ struct EnsembleConfig {
   int max_session_timeout_ms = 60000;
   int latency_ms = 0;               // fixed artificial delay per response
   int election_ms = 0;              // pause after leader kill before failover serves
+  // IO loop threads (0 = auto). Like real ZooKeeper's selector threads, IO
+  // parallelism is independent of ensemble size: connections are assigned
+  // round-robin across the pool regardless of which server accepted them.
+  int io_threads = 0;
   LogLevel log_level = LogLevel::Warn;
 };
 
