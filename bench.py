@@ -167,6 +167,11 @@ def main():
         gathered = [None] * world
         dist.all_gather_object(gathered, rtts_us)
         all_rtts = [r for lst in gathered for r in lst]
+        if os.environ.get("BENCH_DEBUG"):
+            per = [None] * world
+            dist.all_gather_object(per, elapsed)
+            if rank == 0:
+                log("per-rank elapsed: %s" % ", ".join("%.3f" % e for e in per))
     else:
         elapsed_max = elapsed
         all_rtts = rtts_us

@@ -899,6 +899,32 @@ struct Ensemble::Impl {
 
   template <typename BodyFn>
   void send_reply(Conn* c, int32_t xid, int64_t zxid, int32_t err, BodyFn body) {
+    // fast path: serialize straight into the connection's output buffer (one
+    // allocation-free append per reply, flushed once per event drain)
+    if (latency_ms.load(std::memory_order_relaxed) == 0 && loops[c->io_idx]->on_loop_thread()) {
+      {
+        std::lock_guard<std::mutex> og(c->out_mu);
+        if (c->dead.load(std::memory_order_relaxed)) return;
+        size_t start = c->outbuf.size();
+        c->outbuf.append(4, '\0');
+        JuteWriter w(&c->outbuf);
+        ReplyHeader hdr;
+        hdr.xid = xid;
+        hdr.zxid = zxid;
+        hdr.err = err;
+        hdr.serialize(w);
+        if constexpr (!std::is_same_v<BodyFn, std::nullptr_t>) {
+          if (err == kZOk) body(w);
+        }
+        uint32_t n = static_cast<uint32_t>(c->outbuf.size() - start - 4);
+        c->outbuf[start] = static_cast<char>(n >> 24);
+        c->outbuf[start + 1] = static_cast<char>(n >> 16);
+        c->outbuf[start + 2] = static_cast<char>(n >> 8);
+        c->outbuf[start + 3] = static_cast<char>(n);
+      }
+      ensure_flush_scheduled(c);
+      return;
+    }
     std::string pkt;
     begin_packet(&pkt);
     JuteWriter w(&pkt);
@@ -939,19 +965,23 @@ struct Ensemble::Impl {
         if (c->dead.load()) return;
         c->outbuf += pkt;
       }
-      if (!c->flush_scheduled) {
-        c->flush_scheduled = true;
-        uint64_t cid = c->id;
-        loops[c->io_idx]->schedule(0, [this, cid] {
-          ConnPtr cp = lookup(cid);
-          if (!cp) return;
-          cp->flush_scheduled = false;
-          if (!cp->dead.load()) flush_out(cp.get());
-        });
-      }
+      ensure_flush_scheduled(c);
       return;
     }
     deliver(c, std::move(pkt));
+  }
+
+  // owner loop thread only
+  void ensure_flush_scheduled(Conn* c) {
+    if (c->flush_scheduled) return;
+    c->flush_scheduled = true;
+    uint64_t cid = c->id;
+    loops[c->io_idx]->schedule(0, [this, cid] {
+      ConnPtr cp = lookup(cid);
+      if (!cp) return;
+      cp->flush_scheduled = false;
+      if (!cp->dead.load()) flush_out(cp.get());
+    });
   }
 
   // Cross-thread delivery: direct write when the buffer is clear, else

@@ -328,9 +328,10 @@ struct ZkClient::Impl {
         return false;
       }
       if (avail < 4 + len) break;
-      std::string body(inbuf.data() + inpos + 4, len);
+      // parse in place — no per-reply copy; handlers never touch inbuf
+      const char* body = inbuf.data() + inpos + 4;
       inpos += 4 + len;
-      if (!handle_frame(body)) return false;  // handler may reconnect/teardown
+      if (!handle_frame(body, len)) return false;  // handler may reconnect/teardown
     }
     if (inpos > 0 && fd >= 0) {
       inbuf.erase(0, inpos);
@@ -339,9 +340,9 @@ struct ZkClient::Impl {
     return true;
   }
 
-  bool handle_frame(const std::string& body) {
+  bool handle_frame(const char* body, size_t len) {
     try {
-      JuteReader r(body);
+      JuteReader r(body, len);
       if (phase == Phase::Handshaking) {
         ConnectResponse resp;
         resp.deserialize(r);
@@ -445,16 +446,21 @@ struct ZkClient::Impl {
       if (done) done(phase == Phase::Stopped ? kZSessionExpired : kZConnectionLoss, nullptr);
       return;
     }
-    std::string pkt;
-    begin_packet(&pkt);
-    JuteWriter w(&pkt);
+    // serialize straight into the session output buffer (framed in place);
+    // the whole drain cycle flushes as a few large writes
+    size_t start = outbuf.size();
+    outbuf.append(4, '\0');
+    JuteWriter w(&outbuf);
     RequestHeader hdr;
     hdr.xid = xid;
     hdr.type = op;
     hdr.serialize(w);
     serialize_body(w);
-    frame_packet(&pkt);
-    outbuf += pkt;
+    uint32_t n = static_cast<uint32_t>(outbuf.size() - start - 4);
+    outbuf[start] = static_cast<char>(n >> 24);
+    outbuf[start + 1] = static_cast<char>(n >> 16);
+    outbuf[start + 2] = static_cast<char>(n >> 8);
+    outbuf[start + 3] = static_cast<char>(n);
     pending.push_back(Pending{xid, op, std::move(done)});
     schedule_flush();
   }
@@ -800,11 +806,26 @@ int ZkClient::unlink(const std::string& path) { return del(path, -1); }
 // ---- pipelined batches ----
 
 namespace {
+// Blocking batch context: lives on the caller's stack for the whole batch,
+// so callbacks capture a raw pointer + index (fits std::function's SSO — no
+// per-op heap allocation).
 struct BatchState {
   std::mutex mu;
   std::condition_variable cv;
   size_t done = 0;
   size_t total = 0;
+  std::vector<int>* rcs = nullptr;
+  std::vector<Stat>* stats = nullptr;
+
+  void complete(size_t i, int rc) {
+    std::lock_guard<std::mutex> g(mu);
+    (*rcs)[i] = rc;
+    if (++done == total) cv.notify_all();
+  }
+  void wait() {
+    std::unique_lock<std::mutex> g(mu);
+    cv.wait(g, [&] { return done == total; });
+  }
 };
 }  // namespace
 
@@ -813,24 +834,25 @@ std::vector<int> ZkClient::create_many(const std::vector<std::string>& paths, co
   size_t n = paths.size();
   std::vector<int> rcs(n, kZConnectionLoss);
   if (n == 0) return rcs;
-  auto st = std::make_shared<BatchState>();
-  st->total = n;
-  impl_->loop.post([this, &paths, &datas, &rcs, flags, st, n] {
+  BatchState st;
+  st.total = n;
+  st.rcs = &rcs;
+  BatchState* stp = &st;
+  impl_->loop.post([this, &paths, &datas, flags, stp, n] {
     for (size_t i = 0; i < n; i++) {
-      CreateRequest req;
-      req.path = paths[i];
-      req.data = datas[i];
-      req.flags = flags;
-      impl_->submit_op(kOpCreate, [req](JuteWriter& w) { req.serialize(w); },
-                       [st, &rcs, i](int rc, JuteReader*) {
-                         std::lock_guard<std::mutex> g(st->mu);
-                         rcs[i] = rc;
-                         if (++st->done == st->total) st->cv.notify_all();
-                       });
+      const std::string& path = paths[i];
+      const std::string& data = datas[i];
+      impl_->submit_op(kOpCreate,
+                       [&path, &data, flags](JuteWriter& w) {
+                         w.write_string(path);
+                         w.write_buffer(data);
+                         write_acl_vector(w, {ACL{}});
+                         w.write_int(flags);
+                       },
+                       [stp, i](int rc, JuteReader*) { stp->complete(i, rc); });
     }
   });
-  std::unique_lock<std::mutex> g(st->mu);
-  st->cv.wait(g, [&] { return st->done == st->total; });
+  st.wait();
   return rcs;
 }
 
@@ -838,23 +860,22 @@ std::vector<int> ZkClient::delete_many(const std::vector<std::string>& paths) {
   size_t n = paths.size();
   std::vector<int> rcs(n, kZConnectionLoss);
   if (n == 0) return rcs;
-  auto st = std::make_shared<BatchState>();
-  st->total = n;
-  impl_->loop.post([this, &paths, &rcs, st, n] {
+  BatchState st;
+  st.total = n;
+  st.rcs = &rcs;
+  BatchState* stp = &st;
+  impl_->loop.post([this, &paths, stp, n] {
     for (size_t i = 0; i < n; i++) {
-      DeleteRequest req;
-      req.path = paths[i];
-      req.version = -1;
-      impl_->submit_op(kOpDelete, [req](JuteWriter& w) { req.serialize(w); },
-                       [st, &rcs, i](int rc, JuteReader*) {
-                         std::lock_guard<std::mutex> g(st->mu);
-                         rcs[i] = rc;
-                         if (++st->done == st->total) st->cv.notify_all();
-                       });
+      const std::string& path = paths[i];
+      impl_->submit_op(kOpDelete,
+                       [&path](JuteWriter& w) {
+                         w.write_string(path);
+                         w.write_int(-1);
+                       },
+                       [stp, i](int rc, JuteReader*) { stp->complete(i, rc); });
     }
   });
-  std::unique_lock<std::mutex> g(st->mu);
-  st->cv.wait(g, [&] { return st->done == st->total; });
+  st.wait();
   return rcs;
 }
 
@@ -863,28 +884,30 @@ std::vector<int> ZkClient::exists_many(const std::vector<std::string>& paths, st
   std::vector<int> rcs(n, kZConnectionLoss);
   if (stats) stats->assign(n, Stat{});
   if (n == 0) return rcs;
-  auto st = std::make_shared<BatchState>();
-  st->total = n;
-  impl_->loop.post([this, &paths, &rcs, stats, st, n] {
+  BatchState st;
+  st.total = n;
+  st.rcs = &rcs;
+  st.stats = stats;
+  BatchState* stp = &st;
+  impl_->loop.post([this, &paths, stp, n] {
     for (size_t i = 0; i < n; i++) {
-      ExistsRequest req;
-      req.path = paths[i];
-      req.watch = false;
-      impl_->submit_op(kOpExists, [req](JuteWriter& w) { req.serialize(w); },
-                       [st, &rcs, stats, i](int rc, JuteReader* r) {
-                         if (rc == kZOk && r && stats) {
+      const std::string& path = paths[i];
+      impl_->submit_op(kOpExists,
+                       [&path](JuteWriter& w) {
+                         w.write_string(path);
+                         w.write_bool(false);
+                       },
+                       [stp, i](int rc, JuteReader* r) {
+                         if (rc == kZOk && r && stp->stats) {
                            ExistsResponse resp;
                            resp.deserialize(*r);
-                           (*stats)[i] = resp.stat;
+                           (*stp->stats)[i] = resp.stat;
                          }
-                         std::lock_guard<std::mutex> g(st->mu);
-                         rcs[i] = rc;
-                         if (++st->done == st->total) st->cv.notify_all();
+                         stp->complete(i, rc);
                        });
     }
   });
-  std::unique_lock<std::mutex> g(st->mu);
-  st->cv.wait(g, [&] { return st->done == st->total; });
+  st.wait();
   return rcs;
 }
 
