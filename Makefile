@@ -9,7 +9,7 @@ CORE_SRCS := $(SRCDIR)/ensemble.cpp $(SRCDIR)/zkclient.cpp $(SRCDIR)/registrar.c
              $(SRCDIR)/health.cpp $(SRCDIR)/orchestrator.cpp $(SRCDIR)/gpu.cpp
 HDRS := $(wildcard $(SRCDIR)/*.hpp)
 
-.PHONY: all ext daemon test clean
+.PHONY: all ext daemon test clean tsan asan stress
 
 all: ext daemon
 
@@ -28,6 +28,32 @@ $(BINDIR)/zkensembled: $(SRCDIR)/ensemble_main.cpp $(CORE_SRCS) $(HDRS)
 
 test:
 	python3 -m pytest tests/ -x -q -m "not gpu"
+
+# concurrency stress under sanitizers: the multi-threaded core
+# (IO pool, sharded state, cross-thread watch delivery) under chaos
+stress: $(BINDIR)/stress
+	$(BINDIR)/stress -c 4 -t 8
+
+$(BINDIR)/stress: $(SRCDIR)/stress_main.cpp $(CORE_SRCS) $(HDRS)
+	@mkdir -p $(BINDIR)
+	$(CXX) $(CXXFLAGS) -o $@ $(SRCDIR)/stress_main.cpp $(CORE_SRCS)
+
+# gcc-11's libtsan mismodels this glibc's condition_variable timed waits
+# (verified false positives on a textbook cv program), so sanitizer builds
+# use ROCm's LLVM toolchain instead.
+SANCXX ?= /opt/rocm/lib/llvm/bin/clang++
+
+tsan: $(SRCDIR)/stress_main.cpp $(CORE_SRCS) $(HDRS)
+	@mkdir -p $(BINDIR)
+	$(SANCXX) -std=c++17 -O1 -g -fsanitize=thread -pthread -o $(BINDIR)/stress_tsan \
+		$(SRCDIR)/stress_main.cpp $(CORE_SRCS)
+	TSAN_OPTIONS="halt_on_error=1" $(BINDIR)/stress_tsan -c 4 -t 8
+
+asan: $(SRCDIR)/stress_main.cpp $(CORE_SRCS) $(HDRS)
+	@mkdir -p $(BINDIR)
+	$(SANCXX) -std=c++17 -O1 -g -fsanitize=address,undefined -fno-sanitize-recover=all -pthread \
+		-o $(BINDIR)/stress_asan $(SRCDIR)/stress_main.cpp $(CORE_SRCS)
+	$(BINDIR)/stress_asan -c 4 -t 8
 
 clean:
 	rm -rf build bin registrar_amd/*.so

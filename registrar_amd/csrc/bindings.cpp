@@ -219,6 +219,25 @@ PYBIND11_MODULE(_core, m) {
              for (const auto& ev : c.poll_events()) out.append(session_event_to_dict(ev));
              return out;
            })
+      .def("poll_watches",
+           [](zk::ZkClient& c) {
+             py::list out;
+             for (const auto& ev : c.poll_watches()) {
+               py::dict d;
+               const char* type = "none";
+               switch (ev.type) {
+                 case zk::kEventNodeCreated: type = "created"; break;
+                 case zk::kEventNodeDeleted: type = "deleted"; break;
+                 case zk::kEventNodeDataChanged: type = "changed"; break;
+                 case zk::kEventNodeChildrenChanged: type = "child"; break;
+                 default: break;
+               }
+               d["type"] = type;
+               d["path"] = ev.path;
+               out.append(d);
+             }
+             return out;
+           })
       .def("create",
            [](zk::ZkClient& c, const std::string& path, const py::bytes& data, bool ephemeral) {
              std::string d = data;
@@ -234,26 +253,28 @@ PYBIND11_MODULE(_core, m) {
       .def("delete_", &zk::ZkClient::del, py::arg("path"), py::arg("version") = -1,
            py::call_guard<py::gil_scoped_release>())
       .def("exists",
-           [](zk::ZkClient& c, const std::string& path) {
+           [](zk::ZkClient& c, const std::string& path, bool watch) {
              zk::Stat st;
              int rc;
              {
                py::gil_scoped_release rel;
-               rc = c.exists(path, &st);
+               rc = c.exists(path, &st, watch);
              }
              return py::make_tuple(rc, stat_to_dict(st));
-           })
+           },
+           py::arg("path"), py::arg("watch") = false)
       .def("get",
-           [](zk::ZkClient& c, const std::string& path) {
+           [](zk::ZkClient& c, const std::string& path, bool watch) {
              std::string data;
              zk::Stat st;
              int rc;
              {
                py::gil_scoped_release rel;
-               rc = c.get(path, &data, &st);
+               rc = c.get(path, &data, &st, watch);
              }
              return py::make_tuple(rc, py::bytes(data), stat_to_dict(st));
-           })
+           },
+           py::arg("path"), py::arg("watch") = false)
       .def("set",
            [](zk::ZkClient& c, const std::string& path, const py::bytes& data, int version) {
              std::string d = data;
@@ -262,15 +283,16 @@ PYBIND11_MODULE(_core, m) {
            },
            py::arg("path"), py::arg("data"), py::arg("version") = -1)
       .def("get_children",
-           [](zk::ZkClient& c, const std::string& path) {
+           [](zk::ZkClient& c, const std::string& path, bool watch) {
              std::vector<std::string> ch;
              int rc;
              {
                py::gil_scoped_release rel;
-               rc = c.get_children(path, &ch);
+               rc = c.get_children(path, &ch, watch);
              }
              return py::make_tuple(rc, ch);
-           })
+           },
+           py::arg("path"), py::arg("watch") = false)
       .def("put",
            [](zk::ZkClient& c, const std::string& path, const py::bytes& data) {
              std::string d = data;

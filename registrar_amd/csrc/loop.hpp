@@ -12,6 +12,7 @@
 #include <time.h>
 #include <unistd.h>
 
+#include <atomic>
 #include <cstdint>
 #include <cstring>
 #include <deque>
@@ -64,15 +65,23 @@ class EventLoop {
   EventLoop(const EventLoop&) = delete;
   EventLoop& operator=(const EventLoop&) = delete;
 
-  // ---- loop-thread-only API ----
+  // ---- fd registration (thread-safe: a connection may be registered on
+  // this loop by an accepting thread that is not the loop thread) ----
 
   void add_fd(int fd, uint32_t events, FdCallback cb) {
+    {
+      std::lock_guard<std::mutex> g(fd_mu_);
+      fd_cbs_[fd] = std::move(cb);  // callback installed before the fd can fire
+    }
     struct epoll_event ev;
     memset(&ev, 0, sizeof(ev));
     ev.events = events;
     ev.data.fd = fd;
-    if (epoll_ctl(epfd_, EPOLL_CTL_ADD, fd, &ev) < 0) throw std::runtime_error("epoll_ctl ADD failed");
-    fd_cbs_[fd] = std::move(cb);
+    if (epoll_ctl(epfd_, EPOLL_CTL_ADD, fd, &ev) < 0) {
+      std::lock_guard<std::mutex> g(fd_mu_);
+      fd_cbs_.erase(fd);
+      throw std::runtime_error("epoll_ctl ADD failed");
+    }
   }
 
   // Error-tolerant: the fd may already be closed/removed by another thread's
@@ -87,6 +96,7 @@ class EventLoop {
 
   void del_fd(int fd) {
     epoll_ctl(epfd_, EPOLL_CTL_DEL, fd, nullptr);
+    std::lock_guard<std::mutex> g(fd_mu_);
     fd_cbs_.erase(fd);
   }
 
@@ -100,7 +110,7 @@ class EventLoop {
 
   void cancel(TimerId id) { timers_.erase(id); }
 
-  bool running() const { return running_; }
+  bool running() const { return running_.load(std::memory_order_acquire); }
 
   // ---- thread-safe API ----
 
@@ -113,7 +123,7 @@ class EventLoop {
   }
 
   void stop() {
-    post([this] { running_ = false; });
+    post([this] { running_.store(false, std::memory_order_release); });
   }
 
   // Timer arming from any thread (the heap itself is loop-thread-only).
@@ -127,17 +137,17 @@ class EventLoop {
 
   // Run the loop until stop(). Call from the owning thread.
   void run() {
-    running_ = true;
-    loop_tid_ = std::this_thread::get_id();
+    running_.store(true, std::memory_order_release);
+    loop_tid_.store(std::this_thread::get_id(), std::memory_order_release);
     std::vector<struct epoll_event> events(64);
-    while (running_) {
+    while (running_.load(std::memory_order_relaxed)) {
       int timeout = next_timeout_ms();
       int n = epoll_wait(epfd_, events.data(), static_cast<int>(events.size()), timeout);
       if (n < 0) {
         if (errno == EINTR) continue;
         throw std::runtime_error("epoll_wait failed");
       }
-      for (int i = 0; i < n && running_; i++) {
+      for (int i = 0; i < n && running_.load(std::memory_order_relaxed); i++) {
         int fd = events[i].data.fd;
         if (fd == wakeup_fd_) {
           uint64_t v;
@@ -146,12 +156,13 @@ class EventLoop {
           drain_posted();
           continue;
         }
-        auto it = fd_cbs_.find(fd);
-        if (it != fd_cbs_.end()) {
-          // copy: callback may del_fd(fd) and invalidate the map entry
-          FdCallback cb = it->second;
-          cb(events[i].events);
+        FdCallback cb;
+        {
+          std::lock_guard<std::mutex> g(fd_mu_);
+          auto it = fd_cbs_.find(fd);
+          if (it != fd_cbs_.end()) cb = it->second;  // copy: cb may del_fd(fd)
         }
+        if (cb) cb(events[i].events);
       }
       fire_timers();
       if (n == static_cast<int>(events.size())) events.resize(events.size() * 2);
@@ -159,7 +170,9 @@ class EventLoop {
     drain_posted();
   }
 
-  bool on_loop_thread() const { return std::this_thread::get_id() == loop_tid_; }
+  bool on_loop_thread() const {
+    return std::this_thread::get_id() == loop_tid_.load(std::memory_order_acquire);
+  }
 
  private:
   void wakeup() {
@@ -192,7 +205,7 @@ class EventLoop {
 
   void fire_timers() {
     int64_t now = now_ms();
-    while (running_) {
+    while (running_.load(std::memory_order_relaxed)) {
       prune_heap();
       if (heap_.empty() || heap_.top().deadline > now) break;
       TimerId id = heap_.top().id;
@@ -217,8 +230,9 @@ class EventLoop {
 
   int epfd_ = -1;
   int wakeup_fd_ = -1;
-  bool running_ = false;
-  std::thread::id loop_tid_;
+  std::atomic<bool> running_{false};
+  std::atomic<std::thread::id> loop_tid_{};
+  std::mutex fd_mu_;
   std::unordered_map<int, FdCallback> fd_cbs_;
   std::unordered_map<TimerId, std::function<void()>> timers_;
   std::priority_queue<HeapEntry, std::vector<HeapEntry>, std::greater<HeapEntry>> heap_;

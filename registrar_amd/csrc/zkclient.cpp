@@ -93,6 +93,8 @@ struct ZkClient::Impl {
   std::vector<SessionEvent> ev_queue;
   EventCallback ev_cb;
   WatchCallback watch_cb;
+  std::mutex watch_mu;
+  std::vector<WatcherEvent> watch_queue;
   bool settled = false;  // first connect resolved (ok or fail)
 
   Impl(ZkClientConfig c, Logger l) : cfg(std::move(c)), log(l.child("zookeeper")) {
@@ -391,6 +393,10 @@ struct ZkClient::Impl {
         WatcherEvent ev;
         ev.deserialize(r);
         log.debug("watch event", {{"path", Json(ev.path)}, {"type", Json(static_cast<int64_t>(ev.type))}});
+        {
+          std::lock_guard<std::mutex> g(watch_mu);
+          watch_queue.push_back(ev);
+        }
         if (watch_cb) watch_cb(ev);
         return true;
       }
@@ -599,6 +605,13 @@ std::string ZkClient::to_string() const {
 void ZkClient::set_event_callback(EventCallback cb) { impl_->ev_cb = std::move(cb); }
 void ZkClient::set_watch_callback(WatchCallback cb) { impl_->watch_cb = std::move(cb); }
 
+std::vector<WatcherEvent> ZkClient::poll_watches() {
+  std::lock_guard<std::mutex> g(impl_->watch_mu);
+  std::vector<WatcherEvent> out;
+  out.swap(impl_->watch_queue);
+  return out;
+}
+
 std::vector<SessionEvent> ZkClient::poll_events() {
   std::lock_guard<std::mutex> g(impl_->ev_mu);
   std::vector<SessionEvent> out;
@@ -735,17 +748,17 @@ int ZkClient::del(const std::string& path, int32_t version) {
   return p.get_future().get();
 }
 
-int ZkClient::exists(const std::string& path, Stat* stat) {
+int ZkClient::exists(const std::string& path, Stat* stat, bool watch) {
   std::promise<std::pair<int, Stat>> p;
-  aexists(path, false, [&p](int rc, const Stat& st) { p.set_value({rc, st}); });
+  aexists(path, watch, [&p](int rc, const Stat& st) { p.set_value({rc, st}); });
   auto [rc, st] = p.get_future().get();
   if (stat) *stat = st;
   return rc;
 }
 
-int ZkClient::get(const std::string& path, std::string* data, Stat* stat) {
+int ZkClient::get(const std::string& path, std::string* data, Stat* stat, bool watch) {
   std::promise<int> p;
-  aget(path, false, [&](int rc, const std::string& d, const Stat& st) {
+  aget(path, watch, [&](int rc, const std::string& d, const Stat& st) {
     if (data) *data = d;
     if (stat) *stat = st;
     p.set_value(rc);
@@ -762,9 +775,9 @@ int ZkClient::set(const std::string& path, const std::string& data, int32_t vers
   return p.get_future().get();
 }
 
-int ZkClient::get_children(const std::string& path, std::vector<std::string>* children) {
+int ZkClient::get_children(const std::string& path, std::vector<std::string>* children, bool watch) {
   std::promise<int> p;
-  achildren(path, false, [&](int rc, const std::vector<std::string>& ch) {
+  achildren(path, watch, [&](int rc, const std::vector<std::string>& ch) {
     if (children) *children = ch;
     p.set_value(rc);
   });

@@ -129,12 +129,17 @@ class ZkClient {
   void achildren(const std::string& path, bool watch, ChildrenCallback cb);
 
   // ---- sync ops (any thread but the loop thread) ----
+  // watch=true registers a one-shot watch on the path (delivered through the
+  // watch callback and the poll_watches() queue), Binder-style.
   int create(const std::string& path, const std::string& data, int32_t flags, std::string* created_path = nullptr);
   int del(const std::string& path, int32_t version = -1);
-  int exists(const std::string& path, Stat* stat = nullptr);
-  int get(const std::string& path, std::string* data, Stat* stat = nullptr);
+  int exists(const std::string& path, Stat* stat = nullptr, bool watch = false);
+  int get(const std::string& path, std::string* data, Stat* stat = nullptr, bool watch = false);
   int set(const std::string& path, const std::string& data, int32_t version = -1, Stat* stat = nullptr);
-  int get_children(const std::string& path, std::vector<std::string>* children);
+  int get_children(const std::string& path, std::vector<std::string>* children, bool watch = false);
+
+  // drained queue of watch notifications received so far
+  std::vector<WatcherEvent> poll_watches();
 
   // ---- zkplus-surface verbs ----
   // put: create-or-overwrite persistent node (reference lib/register.js:62)
