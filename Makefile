@@ -55,5 +55,15 @@ asan: $(SRCDIR)/stress_main.cpp $(CORE_SRCS) $(HDRS)
 		-o $(BINDIR)/stress_asan $(SRCDIR)/stress_main.cpp $(CORE_SRCS)
 	$(BINDIR)/stress_asan -c 4 -t 8
 
+# release tarball (the reference's `make release`, Makefile:73-92 equivalent)
+VERSION := $(shell python3 -c "import re;print(re.search(r'\"(.*)\"',open('registrar_amd/_version.py').read()).group(1))")
+release: all
+	@mkdir -p dist
+	tar czf dist/mi355x-registrar-$(VERSION).tar.gz \
+		bin/registrard bin/zkensembled etc/ docs/ README.md \
+		registrar_amd/*.py registrar_amd/*.so registrar_amd/csrc \
+		setup.py Makefile
+	@echo "dist/mi355x-registrar-$(VERSION).tar.gz"
+
 clean:
-	rm -rf build bin registrar_amd/*.so
+	rm -rf build bin dist registrar_amd/*.so

@@ -35,17 +35,21 @@ def cmd_daemon(args):
     signal.signal(signal.SIGINT, lambda *_: stop.append(1))
     signal.signal(signal.SIGTERM, lambda *_: stop.append(1))
     orch.start()
+    def drain():
+        for ev in orch.poll_events():
+            print(json.dumps({"event": ev["type"], "detail": ev["detail"], "znodes": ev["znodes"]}),
+                  flush=True)
+
     try:
         while not stop:
-            for ev in orch.poll_events():
-                print(json.dumps({"event": ev["type"], "detail": ev["detail"], "znodes": ev["znodes"]}),
-                      flush=True)
+            drain()
             if orch.expired():
                 print(json.dumps({"event": "sessionExpired", "detail": "exiting (exitOnExpiry)"}), flush=True)
                 return 1
             time.sleep(0.1)
     finally:
         orch.stop()
+        drain()
     return 0
 
 

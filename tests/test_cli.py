@@ -1,0 +1,65 @@
+"""python -m registrar_amd CLI subcommands."""
+import json
+import os
+import signal
+import subprocess
+import sys
+import time
+
+import pytest
+
+from conftest import REPO_ROOT, orch_config, wait_for
+
+ENV = dict(os.environ, PYTHONPATH=REPO_ROOT)
+
+
+def run_cli(*args, **kw):
+    return subprocess.run([sys.executable, "-m", "registrar_amd", *args],
+                          capture_output=True, text=True, env=ENV, timeout=60, **kw)
+
+
+def test_check_valid(tmp_path, ensemble):
+    cfg = orch_config(ensemble, {"domain": "c.test", "type": "host"})
+    p = tmp_path / "cfg.json"
+    p.write_text(json.dumps(cfg))
+    r = run_cli("check", "-f", str(p))
+    assert r.returncode == 0 and "ok" in r.stdout
+
+
+def test_check_invalid(tmp_path):
+    p = tmp_path / "bad.json"
+    p.write_text(json.dumps({"registration": {"domain": "x", "type": "y"}}))  # no zookeeper
+    r = run_cli("check", "-f", str(p))
+    assert r.returncode == 1 and "zookeeper" in r.stderr
+
+
+def test_gpus_runs_anywhere():
+    r = run_cli("gpus")
+    assert r.returncode == 0  # "no GPUs" on CPU boxes is fine
+
+
+def test_tree_dump(ensemble, client):
+    client.mkdirp("/com/example/svc")
+    client.create("/com/example/svc/h1", json.dumps({"type": "host", "address": "1.2.3.4"}).encode(), True)
+    r = run_cli("tree", "--servers", ensemble.connect_string())
+    assert r.returncode == 0
+    assert "h1" in r.stdout and "ephemeral" in r.stdout and "1.2.3.4" in r.stdout
+
+
+def test_daemon_subcommand(ensemble, tmp_path):
+    cfg = orch_config(
+        ensemble,
+        {"domain": "pycli.test", "type": "host", "hostname": "pyh", "settleMs": 0},
+        heartbeatInterval=200,
+    )
+    p = tmp_path / "cfg.json"
+    p.write_text(json.dumps(cfg))
+    proc = subprocess.Popen([sys.executable, "-m", "registrar_amd", "daemon", "-f", str(p)],
+                            stdout=subprocess.PIPE, text=True, env=ENV)
+    try:
+        assert wait_for(lambda: ensemble.get("/test/pycli/pyh")["exists"], timeout=15)
+    finally:
+        proc.send_signal(signal.SIGTERM)
+        out, _ = proc.communicate(timeout=15)
+    assert proc.returncode == 0
+    assert any('"event": "register"' in line for line in out.splitlines())
