@@ -73,7 +73,13 @@ def main():
     if distributed:
         import torch.distributed as dist  # noqa: F811
 
-        backend = os.environ.get("BENCH_BACKEND") or ("nccl" if use_cuda else "gloo")
+        # Coordination barriers only (broadcast of the ensemble address, the
+        # timed-region fence, elapsed/rtt gathers): gloo. The workload itself
+        # is a host-side control plane — registration traffic carries no GPU
+        # tensors and RCCL is deliberately NOT its transport (SURVEY.md §5.8:
+        # xGMI's role here is topology metadata in the payload, not data
+        # movement). BENCH_BACKEND=nccl opts into RCCL barriers instead.
+        backend = os.environ.get("BENCH_BACKEND") or "gloo"
         dist.init_process_group(backend=backend, rank=rank, world_size=world)
 
     def barrier():

@@ -77,7 +77,8 @@ struct Ensemble::Impl {
   struct ZNode {
     std::string data;
     Stat stat;
-    std::set<std::string> children;
+    // unordered: create/delete are the hot path; listings sort on demand
+    std::unordered_set<std::string> children;
   };
 
   struct Shard {
@@ -95,7 +96,7 @@ struct Ensemble::Impl {
     std::atomic<uint64_t> conn_id{0};    // 0 = detached
     std::atomic<bool> alive{true};       // false once kill_session starts
     std::mutex eph_mu;
-    std::set<std::string> ephemerals;
+    std::unordered_set<std::string> ephemerals;
   };
   using SessionPtr = std::shared_ptr<Session>;
 
@@ -420,6 +421,12 @@ struct Ensemble::Impl {
           break;
         case kOpGetChildren:
           handle_get_children(c, hdr.xid, r);
+          break;
+        case kOpGetChildren2:
+          handle_get_children2(c, hdr.xid, r);
+          break;
+        case kOpSync:
+          handle_sync(c, hdr.xid, r);
           break;
         case kOpCloseSession:
           handle_close_session(c, hdr.xid);
@@ -776,6 +783,7 @@ struct Ensemble::Impl {
       } else {
         if (req.watch) sh.child_watches[req.path].insert(sid_of(c));
         resp.children.assign(it->second.children.begin(), it->second.children.end());
+        std::sort(resp.children.begin(), resp.children.end());
       }
     }
     if (err == kZOk)
@@ -783,6 +791,43 @@ struct Ensemble::Impl {
                  [&](JuteWriter& w) { resp.serialize(w); });
     else
       send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), err, nullptr);
+  }
+
+  // getChildren2: children + the node's Stat (3.4+ clients use it by default)
+  void handle_get_children2(Conn* c, int32_t xid, JuteReader& r) {
+    ops.get_children.fetch_add(1, std::memory_order_relaxed);
+    GetChildrenRequest req;
+    req.deserialize(r);
+    int32_t err = kZOk;
+    GetChildrenResponse resp;
+    Stat stat;
+    {
+      Shard& sh = shard_of(req.path);
+      std::lock_guard<std::mutex> lk(sh.mu);
+      auto it = sh.nodes.find(req.path);
+      if (it == sh.nodes.end()) {
+        err = kZNoNode;
+      } else {
+        if (req.watch) sh.child_watches[req.path].insert(sid_of(c));
+        resp.children.assign(it->second.children.begin(), it->second.children.end());
+        std::sort(resp.children.begin(), resp.children.end());
+        stat = it->second.stat;
+      }
+    }
+    if (err == kZOk)
+      send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), kZOk, [&](JuteWriter& w) {
+        resp.serialize(w);
+        stat.serialize(w);
+      });
+    else
+      send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), err, nullptr);
+  }
+
+  // sync: single-copy tree ⇒ always in sync; echo the path back
+  void handle_sync(Conn* c, int32_t xid, JuteReader& r) {
+    std::string path = r.read_string();
+    send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), kZOk,
+               [&](JuteWriter& w) { w.write_string(path); });
   }
 
   void handle_close_session(Conn* c, int32_t xid) {
@@ -812,7 +857,7 @@ struct Ensemble::Impl {
       dead_sessions.insert(sid);
     }
     s->alive.store(false, std::memory_order_release);
-    std::set<std::string> eph;
+    std::unordered_set<std::string> eph;
     {
       std::lock_guard<std::mutex> eg(s->eph_mu);
       eph.swap(s->ephemerals);
@@ -1253,11 +1298,16 @@ NodeInfo Ensemble::get(const std::string& path) const {
 }
 
 std::vector<std::string> Ensemble::children(const std::string& path) const {
-  auto& sh = impl_->shard_of(path);
-  std::lock_guard<std::mutex> g(sh.mu);
-  auto it = sh.nodes.find(path);
-  if (it == sh.nodes.end()) return {};
-  return std::vector<std::string>(it->second.children.begin(), it->second.children.end());
+  std::vector<std::string> out;
+  {
+    auto& sh = impl_->shard_of(path);
+    std::lock_guard<std::mutex> g(sh.mu);
+    auto it = sh.nodes.find(path);
+    if (it == sh.nodes.end()) return out;
+    out.assign(it->second.children.begin(), it->second.children.end());
+  }
+  std::sort(out.begin(), out.end());
+  return out;
 }
 
 size_t Ensemble::node_count() const {
