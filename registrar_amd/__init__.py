@@ -41,6 +41,7 @@ from registrar_amd._core import (  # noqa: F401
     Ensemble,
     HealthCheck,
     Orchestrator,
+    PreparedRegistration,
     ZkClient,
     build_host_record,
     build_node_list,
@@ -72,6 +73,7 @@ __all__ = [
     "Ensemble",
     "HealthCheck",
     "Orchestrator",
+    "PreparedRegistration",
     "ZkClient",
     "build_host_record",
     "build_node_list",

@@ -344,6 +344,30 @@ PYBIND11_MODULE(_core, m) {
         [](const std::string& registration_json) {
           return build_node_list(parse_registration(Json::parse(registration_json)));
         });
+  // Pre-parsed registration: node list / mkdirp chain / payloads derived
+  // once, so repeated registrations (storms, benches) are pure wire work.
+  py::class_<PreparedRegistration>(m, "PreparedRegistration")
+      .def(py::init([](const std::string& registration_json) {
+             return std::make_unique<PreparedRegistration>(
+                 prepare_registration(parse_registration(Json::parse(registration_json))));
+           }),
+           py::arg("registration_json"))
+      .def_property_readonly("path", [](const PreparedRegistration& p) { return p.path; })
+      .def_property_readonly("nodes", [](const PreparedRegistration& p) { return p.nodes; })
+      .def_property_readonly("host_payload",
+                             [](const PreparedRegistration& p) { return py::bytes(p.host_payload); })
+      .def("register_",
+           [](const PreparedRegistration& prep, zk::ZkClient& c, const std::string& log_level) {
+             Logger log = make_logger("registrar", log_level);
+             RegisterResult res;
+             {
+               py::gil_scoped_release rel;
+               res = register_prepared(c, prep, log);
+             }
+             return py::make_tuple(res.rc, res.error, res.znodes);
+           },
+           py::arg("client"), py::arg("log_level") = "warn");
+
   m.def("register_node",
         [](zk::ZkClient& c, const std::string& registration_json, const std::string& log_level) {
           RegistrationConfig cfg = parse_registration(Json::parse(registration_json));

@@ -199,22 +199,40 @@ std::string dirname_of(const std::string& p) {
 }
 }  // namespace
 
-RegisterResult register_node(zk::ZkClient& client, const RegistrationConfig& cfg, const Logger& log) {
-  RegisterResult result;
-  std::string path = domain_to_path(cfg.domain);
-  Logger rlog = log.child({{"component", Json("register")}, {"domain", Json(cfg.domain)}, {"path", Json(path)}});
+PreparedRegistration prepare_registration(const RegistrationConfig& cfg) {
+  PreparedRegistration prep;
+  prep.cfg = cfg;
+  prep.path = domain_to_path(cfg.domain);
+  prep.nodes = build_node_list(cfg);
+  std::set<std::string> prefixes;
+  for (const auto& n : prep.nodes) {
+    std::string dir = dirname_of(n);
+    size_t pos = 0;
+    while ((pos = dir.find('/', pos + 1)) != std::string::npos) prefixes.insert(dir.substr(0, pos));
+    if (dir != "/") prefixes.insert(dir);
+  }
+  // lexicographic order puts every parent before its children
+  prep.dirs.assign(prefixes.begin(), prefixes.end());
+  prep.host_payload = build_host_record(cfg).dump();
+  if (cfg.service) prep.service_payload = build_service_record(cfg).dump();
+  return prep;
+}
 
-  std::vector<std::string> nodes = build_node_list(cfg);
+RegisterResult register_prepared(zk::ZkClient& client, const PreparedRegistration& prep, const Logger& log) {
+  RegisterResult result;
+  const RegistrationConfig& cfg = prep.cfg;
+  Logger rlog =
+      log.child({{"component", Json("register")}, {"domain", Json(cfg.domain)}, {"path", Json(prep.path)}});
   rlog.debug("register: entered");
 
   // 1) cleanupPreviousEntries: parallel unlink, NO_NODE tolerated
   //    (lib/register.js:78-105)
   {
-    std::vector<int> rcs = client.delete_many(nodes);
+    std::vector<int> rcs = client.delete_many(prep.nodes);
     for (size_t i = 0; i < rcs.size(); i++) {
       if (rcs[i] != zk::kZOk && rcs[i] != zk::kZNoNode) {
         result.rc = rcs[i];
-        result.error = std::string("cleanupPreviousEntries: unlink ") + nodes[i] + " failed: " +
+        result.error = std::string("cleanupPreviousEntries: unlink ") + prep.nodes[i] + " failed: " +
                        zk::error_name(rcs[i]);
         rlog.debug("cleanupPreviousEntries: failed", {{"err", Json(result.error)}});
         return result;
@@ -228,24 +246,16 @@ RegisterResult register_node(zk::ZkClient& client, const RegistrationConfig& cfg
   //    here via registration.settleMs)
   if (cfg.settle_ms > 0) std::this_thread::sleep_for(std::chrono::milliseconds(cfg.settle_ms));
 
-  // 3) setupDirectories: mkdirp of every dirname (lib/register.js:108-129).
-  //    All prefix chains are merged, deduped, and pipelined in one flush;
-  //    lexicographic order puts every parent before its children.
+  // 3) setupDirectories: mkdirp of every dirname, prefix chains merged,
+  //    deduped and pipelined in one flush (lib/register.js:108-129)
   {
-    std::set<std::string> prefixes;
-    for (const auto& n : nodes) {
-      std::string dir = dirname_of(n);
-      size_t pos = 0;
-      while ((pos = dir.find('/', pos + 1)) != std::string::npos) prefixes.insert(dir.substr(0, pos));
-      if (dir != "/") prefixes.insert(dir);
-    }
-    std::vector<std::string> dirs(prefixes.begin(), prefixes.end());
-    std::vector<std::string> datas(dirs.size());
-    std::vector<int> rcs = client.create_many(dirs, datas, 0);
+    std::vector<std::string> datas(prep.dirs.size());
+    std::vector<int> rcs = client.create_many(prep.dirs, datas, 0);
     for (size_t i = 0; i < rcs.size(); i++) {
       if (rcs[i] != zk::kZOk && rcs[i] != zk::kZNodeExists) {
         result.rc = rcs[i];
-        result.error = std::string("setupDirectories: mkdirp ") + dirs[i] + " failed: " + zk::error_name(rcs[i]);
+        result.error = std::string("setupDirectories: mkdirp ") + prep.dirs[i] + " failed: " +
+                       zk::error_name(rcs[i]);
         rlog.debug("setupDirectories: failed", {{"err", Json(result.error)}});
         return result;
       }
@@ -256,13 +266,13 @@ RegisterResult register_node(zk::ZkClient& client, const RegistrationConfig& cfg
   // 4) registerEntries: one ephemeral create per node, pipelined
   //    (lib/register.js:132-171 — the regs/sec hot path)
   {
-    std::string payload = build_host_record(cfg).dump();
-    std::vector<std::string> datas(nodes.size(), payload);
-    std::vector<int> rcs = client.create_many(nodes, datas, zk::kEphemeral);
+    std::vector<std::string> datas(prep.nodes.size(), prep.host_payload);
+    std::vector<int> rcs = client.create_many(prep.nodes, datas, zk::kEphemeral);
     for (size_t i = 0; i < rcs.size(); i++) {
       if (rcs[i] != zk::kZOk) {
         result.rc = rcs[i];
-        result.error = std::string("registerEntries: create ") + nodes[i] + " failed: " + zk::error_name(rcs[i]);
+        result.error = std::string("registerEntries: create ") + prep.nodes[i] + " failed: " +
+                       zk::error_name(rcs[i]);
         rlog.debug("registerEntries: failed", {{"err", Json(result.error)}});
         return result;
       }
@@ -270,29 +280,35 @@ RegisterResult register_node(zk::ZkClient& client, const RegistrationConfig& cfg
     rlog.debug("registerEntries: done");
   }
 
+  result.znodes = prep.nodes;
+
   // 5) registerService: persistent put of the service record at $path itself,
   //    appended to the heartbeat node list (lib/register.js:45-75)
   if (cfg.service) {
-    std::string payload = build_service_record(cfg).dump();
-    int rc = client.put(path, payload);
+    int rc = client.put(prep.path, prep.service_payload);
     if (rc != zk::kZOk) {
       result.rc = rc;
-      result.error = std::string("registerService: put ") + path + " failed: " + zk::error_name(rc);
+      result.error = std::string("registerService: put ") + prep.path + " failed: " + zk::error_name(rc);
       rlog.error("registerService: put failed", {{"err", Json(result.error)}});
+      result.znodes.clear();
       return result;
     }
-    if (std::find(nodes.begin(), nodes.end(), path) == nodes.end()) nodes.push_back(path);
+    if (std::find(result.znodes.begin(), result.znodes.end(), prep.path) == result.znodes.end())
+      result.znodes.push_back(prep.path);
     rlog.debug("registerService: done");
   }
 
   result.rc = zk::kZOk;
-  result.znodes = std::move(nodes);
   {
     Json zn = Json::array();
     for (const auto& n : result.znodes) zn.push_back(Json(n));
     rlog.debug("register: done", {{"znodes", std::move(zn)}});
   }
   return result;
+}
+
+RegisterResult register_node(zk::ZkClient& client, const RegistrationConfig& cfg, const Logger& log) {
+  return register_prepared(client, prepare_registration(cfg), log);
 }
 
 int unregister_node(zk::ZkClient& client, const std::vector<std::string>& znodes, const Logger& log) {

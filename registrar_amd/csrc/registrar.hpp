@@ -90,7 +90,23 @@ struct RegisterResult {
   std::vector<std::string> znodes;   // nodes to heartbeat (incl. service path)
 };
 
+// All the derived state a register() needs — node list, deduped mkdirp
+// prefix chain, serialized payloads — computed once so repeated
+// registrations (expiry storms, health recoveries, benchmarks) spend their
+// time on the wire, not re-deriving strings.
+struct PreparedRegistration {
+  RegistrationConfig cfg;
+  std::string path;                  // domainToPath(domain)
+  std::vector<std::string> nodes;    // ephemeral znodes to create
+  std::vector<std::string> dirs;     // sorted unique prefixes (parents first)
+  std::string host_payload;          // serialized host record
+  std::string service_payload;       // serialized service record ("" if none)
+};
+
+PreparedRegistration prepare_registration(const RegistrationConfig& cfg);
+
 // Full 5-step pipeline against a connected client.
+RegisterResult register_prepared(zk::ZkClient& client, const PreparedRegistration& prep, const Logger& log);
 RegisterResult register_node(zk::ZkClient& client, const RegistrationConfig& cfg, const Logger& log);
 
 // Delete every znode; idempotent. Returns zk::kZOk or the first hard error.

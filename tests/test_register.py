@@ -181,3 +181,29 @@ def test_gpu_extension_payload():
     # and without gpu config the key is absent
     rec2 = json.loads(ra.build_host_record(json.dumps({"domain": "a.b", "type": "host", "adminIp": "1.2.3.4"})))
     assert "gpu" not in rec2["host"]
+
+
+def test_prepared_registration_matches_register_node(ensemble, client):
+    import registrar_amd as ra
+
+    registration = {
+        "domain": "prep.test",
+        "type": "host",
+        "adminIp": "127.0.0.1",
+        "hostname": "ph",
+        "settleMs": 0,
+        "aliases": ["a1.prep.test", "a2.prep.test"],
+        "service": SERVICE,
+    }
+    prep = ra.PreparedRegistration(json.dumps(registration))
+    assert prep.path == "/test/prep"
+    assert len(prep.nodes) == 3
+    rc, err, znodes = prep.register_(client)
+    assert rc == ra.ZOK, err
+    assert "/test/prep" in znodes  # service path joins the heartbeat list
+    # same observable result as the one-shot engine call
+    obj = json.loads(ensemble.get("/test/prep/ph")["data"])
+    assert obj == json.loads(prep.host_payload)
+    # idempotent re-register through the same prepared handle
+    rc, err, znodes2 = prep.register_(client)
+    assert rc == ra.ZOK and znodes2 == znodes
