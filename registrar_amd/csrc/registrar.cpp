@@ -215,6 +215,27 @@ PreparedRegistration prepare_registration(const RegistrationConfig& cfg) {
   prep.dirs.assign(prefixes.begin(), prefixes.end());
   prep.host_payload = build_host_record(cfg).dump();
   if (cfg.service) prep.service_payload = build_service_record(cfg).dump();
+  prep.wire_ops.reserve(prep.nodes.size() * 2 + prep.dirs.size());
+  for (const auto& n : prep.nodes) {
+    zk::ZkClient::MixedOp op;
+    op.op = zk::kOpDelete;
+    op.path = n;
+    prep.wire_ops.push_back(std::move(op));
+  }
+  for (const auto& d : prep.dirs) {
+    zk::ZkClient::MixedOp op;
+    op.op = zk::kOpCreate;
+    op.path = d;
+    prep.wire_ops.push_back(std::move(op));
+  }
+  for (const auto& n : prep.nodes) {
+    zk::ZkClient::MixedOp op;
+    op.op = zk::kOpCreate;
+    op.path = n;
+    op.data = prep.host_payload;
+    op.flags = zk::kEphemeral;
+    prep.wire_ops.push_back(std::move(op));
+  }
   return prep;
 }
 
@@ -225,62 +246,77 @@ RegisterResult register_prepared(zk::ZkClient& client, const PreparedRegistratio
       log.child({{"component", Json("register")}, {"domain", Json(cfg.domain)}, {"path", Json(prep.path)}});
   rlog.debug("register: entered");
 
-  // 1) cleanupPreviousEntries: parallel unlink, NO_NODE tolerated
-  //    (lib/register.js:78-105)
-  {
+  size_t n_nodes = prep.nodes.size();
+  size_t n_dirs = prep.dirs.size();
+
+  auto check_cleanup = [&](const std::vector<int>& rcs, size_t base) -> bool {
+    // cleanupPreviousEntries: NO_NODE tolerated (lib/register.js:78-105)
+    for (size_t i = 0; i < n_nodes; i++) {
+      int rc = rcs[base + i];
+      if (rc != zk::kZOk && rc != zk::kZNoNode) {
+        result.rc = rc;
+        result.error =
+            std::string("cleanupPreviousEntries: unlink ") + prep.nodes[i] + " failed: " + zk::error_name(rc);
+        return false;
+      }
+    }
+    return true;
+  };
+  auto check_dirs = [&](const std::vector<int>& rcs, size_t base) -> bool {
+    // setupDirectories: NODE_EXISTS tolerated (lib/register.js:108-129)
+    for (size_t i = 0; i < n_dirs; i++) {
+      int rc = rcs[base + i];
+      if (rc != zk::kZOk && rc != zk::kZNodeExists) {
+        result.rc = rc;
+        result.error = std::string("setupDirectories: mkdirp ") + prep.dirs[i] + " failed: " + zk::error_name(rc);
+        return false;
+      }
+    }
+    return true;
+  };
+  auto check_creates = [&](const std::vector<int>& rcs, size_t base) -> bool {
+    // registerEntries: every ephemeral create must succeed
+    for (size_t i = 0; i < n_nodes; i++) {
+      int rc = rcs[base + i];
+      if (rc != zk::kZOk) {
+        result.rc = rc;
+        result.error =
+            std::string("registerEntries: create ") + prep.nodes[i] + " failed: " + zk::error_name(rc);
+        return false;
+      }
+    }
+    return true;
+  };
+
+  if (cfg.settle_ms > 0) {
+    // settle configured: a real barrier after cleanup (reference semantics,
+    // fixed 1000 ms at lib/register.js:232-235), then the remaining stages
+    // in one pipelined round trip
     std::vector<int> rcs = client.delete_many(prep.nodes);
-    for (size_t i = 0; i < rcs.size(); i++) {
-      if (rcs[i] != zk::kZOk && rcs[i] != zk::kZNoNode) {
-        result.rc = rcs[i];
-        result.error = std::string("cleanupPreviousEntries: unlink ") + prep.nodes[i] + " failed: " +
-                       zk::error_name(rcs[i]);
-        rlog.debug("cleanupPreviousEntries: failed", {{"err", Json(result.error)}});
-        return result;
-      }
+    if (!check_cleanup(rcs, 0)) {
+      rlog.debug("cleanupPreviousEntries: failed", {{"err", Json(result.error)}});
+      return result;
     }
-    rlog.debug("cleanupPreviousEntries: done");
+    std::this_thread::sleep_for(std::chrono::milliseconds(cfg.settle_ms));
+    std::vector<zk::ZkClient::MixedOp> rest(prep.wire_ops.begin() + static_cast<long>(n_nodes),
+                                            prep.wire_ops.end());
+    std::vector<int> rcs2 = client.submit_mixed(rest);
+    if (!check_dirs(rcs2, 0) || !check_creates(rcs2, n_dirs)) {
+      rlog.debug("register: failed", {{"err", Json(result.error)}});
+      return result;
+    }
+  } else {
+    // no settle: the whole cleanup → mkdirp → create sequence is one
+    // pipelined submission — ZooKeeper's per-session in-order processing
+    // guarantees the same final state as the reference's staged barriers
+    std::vector<int> rcs = client.submit_mixed(prep.wire_ops);
+    if (!check_cleanup(rcs, 0) || !check_dirs(rcs, n_nodes) || !check_creates(rcs, n_nodes + n_dirs)) {
+      rlog.debug("register: failed", {{"err", Json(result.error)}});
+      return result;
+    }
   }
 
-  // 2) settle: be nice to watchers and wait for them to catch up
-  //    (fixed 1000 ms in the reference, lib/register.js:232-235; configurable
-  //    here via registration.settleMs)
-  if (cfg.settle_ms > 0) std::this_thread::sleep_for(std::chrono::milliseconds(cfg.settle_ms));
-
-  // 3) setupDirectories: mkdirp of every dirname, prefix chains merged,
-  //    deduped and pipelined in one flush (lib/register.js:108-129)
-  {
-    std::vector<std::string> datas(prep.dirs.size());
-    std::vector<int> rcs = client.create_many(prep.dirs, datas, 0);
-    for (size_t i = 0; i < rcs.size(); i++) {
-      if (rcs[i] != zk::kZOk && rcs[i] != zk::kZNodeExists) {
-        result.rc = rcs[i];
-        result.error = std::string("setupDirectories: mkdirp ") + prep.dirs[i] + " failed: " +
-                       zk::error_name(rcs[i]);
-        rlog.debug("setupDirectories: failed", {{"err", Json(result.error)}});
-        return result;
-      }
-    }
-    rlog.debug("setupDirectories: done");
-  }
-
-  // 4) registerEntries: one ephemeral create per node, pipelined
-  //    (lib/register.js:132-171 — the regs/sec hot path)
-  {
-    std::vector<std::string> datas(prep.nodes.size(), prep.host_payload);
-    std::vector<int> rcs = client.create_many(prep.nodes, datas, zk::kEphemeral);
-    for (size_t i = 0; i < rcs.size(); i++) {
-      if (rcs[i] != zk::kZOk) {
-        result.rc = rcs[i];
-        result.error = std::string("registerEntries: create ") + prep.nodes[i] + " failed: " +
-                       zk::error_name(rcs[i]);
-        rlog.debug("registerEntries: failed", {{"err", Json(result.error)}});
-        return result;
-      }
-    }
-    rlog.debug("registerEntries: done");
-  }
-
-  result.znodes = prep.nodes;
+    result.znodes = prep.nodes;
 
   // 5) registerService: persistent put of the service record at $path itself,
   //    appended to the heartbeat node list (lib/register.js:45-75)

@@ -101,6 +101,10 @@ struct PreparedRegistration {
   std::vector<std::string> dirs;     // sorted unique prefixes (parents first)
   std::string host_payload;          // serialized host record
   std::string service_payload;       // serialized service record ("" if none)
+  // the full wire sequence (cleanup deletes, dir creates, ephemeral
+  // creates) built once; with settle_ms == 0 it ships as a single pipelined
+  // round trip (see zkclient submit_mixed)
+  std::vector<zk::ZkClient::MixedOp> wire_ops;
 };
 
 PreparedRegistration prepare_registration(const RegistrationConfig& cfg);

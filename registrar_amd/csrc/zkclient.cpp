@@ -924,6 +924,40 @@ std::vector<int> ZkClient::exists_many(const std::vector<std::string>& paths, st
   return rcs;
 }
 
+std::vector<int> ZkClient::submit_mixed(const std::vector<MixedOp>& ops) {
+  size_t n = ops.size();
+  std::vector<int> rcs(n, kZConnectionLoss);
+  if (n == 0) return rcs;
+  BatchState st;
+  st.total = n;
+  st.rcs = &rcs;
+  BatchState* stp = &st;
+  impl_->loop.post([this, &ops, stp, n] {
+    for (size_t i = 0; i < n; i++) {
+      const MixedOp& mo = ops[i];
+      if (mo.op == kOpDelete) {
+        impl_->submit_op(kOpDelete,
+                         [&mo](JuteWriter& w) {
+                           w.write_string(mo.path);
+                           w.write_int(-1);
+                         },
+                         [stp, i](int rc, JuteReader*) { stp->complete(i, rc); });
+      } else {
+        impl_->submit_op(kOpCreate,
+                         [&mo](JuteWriter& w) {
+                           w.write_string(mo.path);
+                           w.write_buffer(mo.data);
+                           write_acl_vector(w, {ACL{}});
+                           w.write_int(mo.flags);
+                         },
+                         [stp, i](int rc, JuteReader*) { stp->complete(i, rc); });
+      }
+    }
+  });
+  st.wait();
+  return rcs;
+}
+
 int ZkClient::heartbeat(const std::vector<std::string>& nodes, const RetryPolicy& retry, int64_t* rtt_us) {
   Backoff bo;
   bo.initial_ms = retry.initial_delay_ms;

@@ -156,6 +156,19 @@ class ZkClient {
   std::vector<int> delete_many(const std::vector<std::string>& paths);
   std::vector<int> exists_many(const std::vector<std::string>& paths, std::vector<Stat>* stats = nullptr);
 
+  // Heterogeneous pipelined batch (delete/create mixes). ZooKeeper processes
+  // a session's requests strictly in order, so a stage sequence whose only
+  // dependency is that ordering (cleanup unlinks → parent creates →
+  // ephemeral creates) can ship as ONE submission — one round trip instead
+  // of one per stage.
+  struct MixedOp {
+    int32_t op = kOpCreate;  // kOpCreate or kOpDelete
+    std::string path;
+    std::string data;        // create only
+    int32_t flags = 0;       // create only
+  };
+  std::vector<int> submit_mixed(const std::vector<MixedOp>& ops);
+
   // App-level heartbeat: parallel exists over `nodes` with bounded retry
   // (reference lib/zk.js:21-44: ≤5 attempts, 1 s → 30 s). Returns kZOk when a
   // round succeeded for every node; fills rtt_us with the successful round's
