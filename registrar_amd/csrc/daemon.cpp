@@ -15,6 +15,7 @@
 #include <signal.h>
 #include <unistd.h>
 
+#include <algorithm>
 #include <atomic>
 #include <chrono>
 #include <cstdio>
@@ -31,7 +32,9 @@ using namespace registrar;
 namespace {
 
 std::atomic<int> g_signal{0};
+std::atomic<bool> g_dump_metrics{false};
 void on_signal(int sig) { g_signal.store(sig); }
+void on_usr1(int) { g_dump_metrics.store(true); }
 
 void usage(const char* argv0, const char* msg) {
   if (msg) fprintf(stderr, "%s\n", msg);
@@ -126,6 +129,7 @@ int main(int argc, char** argv) {
 
   signal(SIGINT, on_signal);
   signal(SIGTERM, on_signal);
+  signal(SIGUSR1, on_usr1);  // operational metrics dump
   signal(SIGPIPE, SIG_IGN);
 
   Orchestrator orch(ocfg, log);
@@ -185,6 +189,23 @@ int main(int argc, char** argv) {
     if (orch.expired()) {
       exit_code.store(1);
       break;
+    }
+    if (g_dump_metrics.exchange(false)) {
+      OrchMetrics mtx = orch.metrics();
+      int64_t p50 = 0;
+      if (!mtx.recent_heartbeat_rtt_us.empty()) {
+        std::vector<int64_t> v = mtx.recent_heartbeat_rtt_us;
+        std::nth_element(v.begin(), v.begin() + static_cast<long>(v.size() / 2), v.end());
+        p50 = v[v.size() / 2];
+      }
+      log.info("registrar: metrics",
+               {{"registers", Json(static_cast<int64_t>(mtx.registers))},
+                {"unregisters", Json(static_cast<int64_t>(mtx.unregisters))},
+                {"heartbeats", Json(static_cast<int64_t>(mtx.heartbeats))},
+                {"heartbeatFailures", Json(static_cast<int64_t>(mtx.heartbeat_failures))},
+                {"sessionExpiries", Json(static_cast<int64_t>(mtx.session_expiries))},
+                {"errors", Json(static_cast<int64_t>(mtx.errors))},
+                {"p50HeartbeatRttUs", Json(p50)}});
     }
     std::this_thread::sleep_for(std::chrono::milliseconds(100));
   }

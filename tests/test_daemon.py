@@ -125,3 +125,26 @@ def test_daemon_usage():
     proc = subprocess.run([daemon, "-h"], capture_output=True, text=True, timeout=10)
     assert proc.returncode == 0
     assert "usage:" in proc.stdout
+
+
+def test_daemon_sigusr1_metrics(ensemble, daemon_bin):
+    cfg = orch_config(
+        ensemble,
+        {"domain": "m.test", "type": "host", "hostname": "mh", "settleMs": 0},
+        heartbeatInterval=100,
+    )
+    proc, cfg_path = spawn_daemon(daemon_bin, cfg)
+    try:
+        assert wait_for(lambda: ensemble.get("/test/m/mh")["exists"], timeout=10)
+        time.sleep(0.5)
+        proc.send_signal(signal.SIGUSR1)
+        time.sleep(0.5)
+    finally:
+        proc.send_signal(signal.SIGTERM)
+        recs = read_logs(proc)
+        os.unlink(cfg_path)
+    metrics = [r for r in recs if r["msg"] == "registrar: metrics"]
+    assert metrics, "no metrics record after SIGUSR1"
+    assert metrics[0]["registers"] >= 1
+    assert metrics[0]["heartbeats"] >= 1
+    assert metrics[0]["p50HeartbeatRttUs"] > 0
