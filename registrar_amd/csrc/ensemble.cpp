@@ -428,6 +428,9 @@ struct Ensemble::Impl {
         case kOpSync:
           handle_sync(c, hdr.xid, r);
           break;
+        case kOpSetWatches:
+          handle_set_watches(c, hdr.xid, r);
+          break;
         case kOpCloseSession:
           handle_close_session(c, hdr.xid);
           break;
@@ -821,6 +824,60 @@ struct Ensemble::Impl {
       });
     else
       send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), err, nullptr);
+  }
+
+  // setWatches: re-arm a reconnected session's watches, firing synthetic
+  // events for anything that changed past relative_zxid while it was away
+  void handle_set_watches(Conn* c, int32_t xid, JuteReader& r) {
+    SetWatchesRequest req;
+    req.deserialize(r);
+    int64_t sid = sid_of(c);
+    for (const auto& path : req.data_watches) {
+      bool fire_changed = false, fire_deleted = false;
+      {
+        Shard& sh = shard_of(path);
+        std::lock_guard<std::mutex> lk(sh.mu);
+        auto it = sh.nodes.find(path);
+        if (it == sh.nodes.end())
+          fire_deleted = true;
+        else if (it->second.stat.mzxid > req.relative_zxid)
+          fire_changed = true;
+        else
+          sh.data_watches[path].insert(sid);
+      }
+      if (fire_deleted) deliver_watch({sid}, path, kEventNodeDeleted);
+      if (fire_changed) deliver_watch({sid}, path, kEventNodeDataChanged);
+    }
+    for (const auto& path : req.exist_watches) {
+      bool fire_created = false;
+      {
+        Shard& sh = shard_of(path);
+        std::lock_guard<std::mutex> lk(sh.mu);
+        auto it = sh.nodes.find(path);
+        if (it != sh.nodes.end())
+          fire_created = true;  // appeared while the client was away
+        else
+          sh.data_watches[path].insert(sid);
+      }
+      if (fire_created) deliver_watch({sid}, path, kEventNodeCreated);
+    }
+    for (const auto& path : req.child_watches) {
+      bool fire_child = false, fire_deleted = false;
+      {
+        Shard& sh = shard_of(path);
+        std::lock_guard<std::mutex> lk(sh.mu);
+        auto it = sh.nodes.find(path);
+        if (it == sh.nodes.end())
+          fire_deleted = true;
+        else if (it->second.stat.pzxid > req.relative_zxid)
+          fire_child = true;
+        else
+          sh.child_watches[path].insert(sid);
+      }
+      if (fire_deleted) deliver_watch({sid}, path, kEventNodeDeleted);
+      if (fire_child) deliver_watch({sid}, path, kEventNodeChildrenChanged);
+    }
+    send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), kZOk, nullptr);
   }
 
   // sync: single-copy tree ⇒ always in sync; echo the path back

@@ -519,6 +519,40 @@ struct GetChildrenResponse {
   }
 };
 
+// setWatches (op 101, xid -8): re-arm watches after a same-session
+// reconnect; the server fires synthetic events for changes that happened
+// after relative_zxid while the client was disconnected.
+struct SetWatchesRequest {
+  int64_t relative_zxid = 0;
+  std::vector<std::string> data_watches;
+  std::vector<std::string> exist_watches;
+  std::vector<std::string> child_watches;
+
+  static void write_vec(JuteWriter& w, const std::vector<std::string>& v) {
+    w.write_int(static_cast<int32_t>(v.size()));
+    for (const auto& s : v) w.write_string(s);
+  }
+  static std::vector<std::string> read_vec(JuteReader& r) {
+    int32_t n = r.read_int();
+    std::vector<std::string> v;
+    for (int32_t i = 0; i < n; i++) v.push_back(r.read_string());
+    return v;
+  }
+
+  void serialize(JuteWriter& w) const {
+    w.write_long(relative_zxid);
+    write_vec(w, data_watches);
+    write_vec(w, exist_watches);
+    write_vec(w, child_watches);
+  }
+  void deserialize(JuteReader& r) {
+    relative_zxid = r.read_long();
+    data_watches = read_vec(r);
+    exist_watches = read_vec(r);
+    child_watches = read_vec(r);
+  }
+};
+
 struct WatcherEvent {
   int32_t type = kEventNone;
   int32_t state = kStateSyncConnected;

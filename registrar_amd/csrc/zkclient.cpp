@@ -13,6 +13,7 @@
 #include <chrono>
 #include <future>
 #include <thread>
+#include <unordered_set>
 
 namespace registrar {
 namespace zk {
@@ -83,6 +84,11 @@ struct ZkClient::Impl {
   EventLoop::TimerId ping_timer = 0;
   EventLoop::TimerId retry_timer = 0;
   std::string current_server;
+  // armed one-shot watches (loop thread); re-sent via setWatches after a
+  // same-session reconnect so watches survive connection loss
+  std::unordered_set<std::string> data_watch_paths;
+  std::unordered_set<std::string> exist_watch_paths;
+  std::unordered_set<std::string> child_watch_paths;
 
   // ---- shared state ----
   std::atomic<SessionState> state{SessionState::Connecting};
@@ -357,6 +363,9 @@ struct ZkClient::Impl {
             log.warn("zookeeper: session expired by server", {{"session", Json(sid)}});
             teardown_socket();
             fail_all_pending(kZSessionExpired);
+            data_watch_paths.clear();
+            exist_watch_paths.clear();
+            child_watch_paths.clear();
             phase = Phase::Stopped;
             state.store(SessionState::Expired);
             emit({SessionEvent::Type::Expired, sid, 0, 0});
@@ -376,6 +385,17 @@ struct ZkClient::Impl {
         connect_backoff.reset();
         reconnect_backoff.reset();
         last_recv = now_ms();
+        if (reconnected &&
+            (!data_watch_paths.empty() || !exist_watch_paths.empty() || !child_watch_paths.empty())) {
+          // re-arm watches; the server fires synthetic events for changes
+          // missed past last_zxid (ZooKeeper setWatches semantics)
+          SetWatchesRequest sw;
+          sw.relative_zxid = last_zxid;
+          sw.data_watches.assign(data_watch_paths.begin(), data_watch_paths.end());
+          sw.exist_watches.assign(exist_watch_paths.begin(), exist_watch_paths.end());
+          sw.child_watches.assign(child_watch_paths.begin(), child_watch_paths.end());
+          submit(kXidSetWatches, kOpSetWatches, [sw](JuteWriter& w) { sw.serialize(w); }, nullptr);
+        }
         schedule_ping();
         state.store(SessionState::Connected);
         log.info(reconnected ? "zookeeper: reconnected" : "ZK: connected",
@@ -393,6 +413,9 @@ struct ZkClient::Impl {
         WatcherEvent ev;
         ev.deserialize(r);
         log.debug("watch event", {{"path", Json(ev.path)}, {"type", Json(static_cast<int64_t>(ev.type))}});
+        data_watch_paths.erase(ev.path);
+        exist_watch_paths.erase(ev.path);
+        child_watch_paths.erase(ev.path);
         {
           std::lock_guard<std::mutex> g(watch_mu);
           watch_queue.push_back(ev);
@@ -658,8 +681,17 @@ void ZkClient::aexists(const std::string& path, bool watch, StatCallback cb) {
     ExistsRequest req;
     req.path = path;
     req.watch = watch;
+    Impl* impl = impl_.get();
     impl_->submit_op(kOpExists, [req](JuteWriter& w) { req.serialize(w); },
-                     [cb](int rc, JuteReader* r) {
+                     [cb, impl, req](int rc, JuteReader* r) {
+                       if (req.watch) {
+                         // exists registers a watch on present AND absent
+                         // nodes (created-event semantics)
+                         if (rc == kZOk)
+                           impl->data_watch_paths.insert(req.path);
+                         else if (rc == kZNoNode)
+                           impl->exist_watch_paths.insert(req.path);
+                       }
                        if (!cb) return;
                        Stat st;
                        if (rc == kZOk && r) {
@@ -677,8 +709,10 @@ void ZkClient::aget(const std::string& path, bool watch, DataCallback cb) {
     GetDataRequest req;
     req.path = path;
     req.watch = watch;
+    Impl* impl = impl_.get();
     impl_->submit_op(kOpGetData, [req](JuteWriter& w) { req.serialize(w); },
-                     [cb](int rc, JuteReader* r) {
+                     [cb, impl, req](int rc, JuteReader* r) {
+                       if (req.watch && rc == kZOk) impl->data_watch_paths.insert(req.path);
                        if (!cb) return;
                        std::string data;
                        Stat st;
@@ -718,8 +752,10 @@ void ZkClient::achildren(const std::string& path, bool watch, ChildrenCallback c
     GetChildrenRequest req;
     req.path = path;
     req.watch = watch;
+    Impl* impl = impl_.get();
     impl_->submit_op(kOpGetChildren, [req](JuteWriter& w) { req.serialize(w); },
-                     [cb](int rc, JuteReader* r) {
+                     [cb, impl, req](int rc, JuteReader* r) {
+                       if (req.watch && rc == kZOk) impl->child_watch_paths.insert(req.path);
                        if (!cb) return;
                        std::vector<std::string> children;
                        if (rc == kZOk && r) {

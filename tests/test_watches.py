@@ -96,3 +96,41 @@ def test_watch_survives_across_requests(ensemble):
     assert not any(w["path"] == "/iso" for w in d2())
     c1.close()
     c2.close()
+
+
+def test_watches_survive_reconnect(ensemble3):
+    """setWatches: a same-session reconnect re-arms watches; changes made
+    while disconnected fire synthetic events on reconnect."""
+    c = make_client(ensemble3)
+    other = make_client(ensemble3)
+    drain = watches(c)
+
+    other.mkdirp("/sw")
+    other.create("/sw/live", b"v0")
+    c.get("/sw/live", watch=True)       # data watch
+    rc, _ = c.exists("/sw/coming", watch=True)  # exist watch on missing node
+    assert rc == ra.ZNONODE
+    c.get_children("/sw", watch=True)   # child watch
+
+    # find and kill the server c is attached to: kill servers until c drops,
+    # but keep one up; do the changes while c is reconnecting
+    sid = c.session_id()
+    ensemble3.kill_server(0)
+    ensemble3.kill_server(1)
+    other2 = None
+    # `other` may have been on a killed server too; use a fresh client for
+    # the mutations (server 2 is still up)
+    other.close()
+    other2 = make_client(ensemble3)
+    other2.set("/sw/live", b"v1")           # missed data change
+    other2.create("/sw/coming", b"here")    # missed creation (child of /sw too)
+
+    # c reconnects (same session) to server 2 and must see all three events
+    assert wait_for(lambda: c.session_id() == sid and c.state() == "connected", timeout=10)
+    assert wait_for(lambda: any(w["type"] == "changed" and w["path"] == "/sw/live" for w in drain()), 10)
+    assert wait_for(lambda: any(w["type"] == "created" and w["path"] == "/sw/coming" for w in drain()), 10)
+    assert wait_for(lambda: any(w["type"] == "child" and w["path"] == "/sw" for w in drain()), 10)
+    other2.close()
+    c.close()
+    ensemble3.restart_server(0)
+    ensemble3.restart_server(1)
