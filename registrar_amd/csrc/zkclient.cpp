@@ -152,12 +152,18 @@ struct ZkClient::Impl {
     addr.sin_family = AF_INET;
     addr.sin_port = htons(static_cast<uint16_t>(srv.port));
     if (inet_pton(AF_INET, srv.host.c_str(), &addr.sin_addr) != 1) {
-      struct hostent* he = gethostbyname(srv.host.c_str());
-      if (!he || he->h_addrtype != AF_INET) {
+      struct addrinfo hints;
+      memset(&hints, 0, sizeof(hints));
+      hints.ai_family = AF_INET;
+      hints.ai_socktype = SOCK_STREAM;
+      struct addrinfo* res = nullptr;
+      if (getaddrinfo(srv.host.c_str(), nullptr, &hints, &res) != 0 || !res) {
+        if (res) freeaddrinfo(res);
         on_connect_failed("resolve failed");
         return;
       }
-      memcpy(&addr.sin_addr, he->h_addr_list[0], sizeof(addr.sin_addr));
+      addr.sin_addr = reinterpret_cast<struct sockaddr_in*>(res->ai_addr)->sin_addr;
+      freeaddrinfo(res);
     }
 
     fd = socket(AF_INET, SOCK_STREAM | SOCK_CLOEXEC, 0);
@@ -1020,7 +1026,14 @@ int ZkClient::heartbeat(const std::vector<std::string>& nodes, const RetryPolicy
     if (bo.exhausted()) return rc;
     int64_t delay = bo.next_delay();
     impl_->log.debug("heartbeat: retrying", {{"rc", Json(error_name(rc))}, {"delay_ms", Json(delay)}});
-    std::this_thread::sleep_for(std::chrono::milliseconds(delay));
+    // interruptible: close()/expiry wake ev_cv, so shutdown never waits out
+    // a long backoff
+    std::unique_lock<std::mutex> g(impl_->ev_mu);
+    impl_->ev_cv.wait_for(g, std::chrono::milliseconds(delay), [this] {
+      SessionState st = impl_->state.load();
+      return impl_->closed.load() || st == SessionState::Expired || st == SessionState::Closed;
+    });
+    if (impl_->closed.load()) return kZConnectionLoss;
   }
 }
 

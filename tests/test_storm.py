@@ -126,3 +126,31 @@ def test_aliases_multipath_flap(ensemble):
     assert wait_for(lambda: all(ensemble.get(n)["exists"] for n in znodes), timeout=10)
     o.stop()
     os.rmdir(tmpdir)
+
+
+def test_register_10k_znodes_scale(ensemble, client):
+    """10x the BASELINE scale in one registration: exercises buffer growth,
+    shard distribution, and large pipelined batches."""
+    registration = {
+        "domain": "big.scale.test",
+        "type": "host",
+        "adminIp": "127.0.0.1",
+        "hostname": "big0",
+        "settleMs": 0,
+        "aliases": ["a%05d.big.scale.test" % i for i in range(9999)],
+    }
+    import registrar_amd as ra
+
+    prep = ra.PreparedRegistration(json.dumps(registration))
+    t0 = time.monotonic()
+    rc, err, znodes = prep.register_(client)
+    dt = time.monotonic() - t0
+    assert rc == ra.ZOK, err
+    assert len(znodes) == 10000
+    assert ensemble.ephemeral_count() == 10000
+    assert dt < 30.0, "10k-node register took %.2fs" % dt
+    rc, rtt_us = client.heartbeat(znodes)
+    assert rc == ra.ZOK
+    assert rtt_us < 3_000_000  # still inside the heartbeat cadence envelope
+    rcs = client.delete_many(znodes)
+    assert all(r == ra.ZOK for r in rcs)
