@@ -42,23 +42,26 @@ def _worker(connect, idx, q):
     c.close()
 
 
-def test_four_processes_share_ensemble(ensemble):
+@pytest.mark.parametrize("nprocs", [4, 8])
+def test_processes_share_ensemble(ensemble3, nprocs):
+    # the 8-process axis of the north star, hermetically on CPU (the GPU
+    # bench covers the same shape with timing)
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    connect = ensemble.connect_string()
-    procs = [ctx.Process(target=_worker, args=(connect, i, q)) for i in range(4)]
+    connect = ensemble3.connect_string()
+    procs = [ctx.Process(target=_worker, args=(connect, i, q)) for i in range(nprocs)]
     for p in procs:
         p.start()
-    results = [q.get(timeout=60) for _ in procs]
+    results = [q.get(timeout=120) for _ in procs]
     for p in procs:
-        p.join(timeout=30)
+        p.join(timeout=60)
         assert p.exitcode == 0
     assert all(status == "ok" for _, status, _ in results), results
     all_nodes = [n for _, _, zn in results for n in zn]
-    assert len(all_nodes) == 40
-    assert len(set(all_nodes)) == 40  # no collisions across processes
+    assert len(all_nodes) == nprocs * 10
+    assert len(set(all_nodes)) == nprocs * 10  # no collisions across processes
     # processes have exited ⇒ sessions close ⇒ ephemerals vanish
-    assert wait_for(lambda: ensemble.ephemeral_count() == 0, timeout=15)
+    assert wait_for(lambda: ensemble3.ephemeral_count() == 0, timeout=20)
 
 
 def _dist_worker(rank, world, port, connect, q):

@@ -170,3 +170,41 @@ def test_adminip_hoist(ensemble):
     obj = json.loads(ensemble.get(o.znodes()[0])["data"])
     assert obj["address"] == "10.9.9.9"
     o.stop()
+
+
+def test_config2_load_balancer_srv(ensemble3):
+    """BASELINE config 2: load_balancer with SRV ports, 3-node ensemble,
+    /bin/true health exec (fast cadence here)."""
+    registration = {
+        "domain": "lb.us-east.example.com",
+        "type": "load_balancer",
+        "adminIp": "127.0.0.1",
+        "hostname": "lb0",
+        "settleMs": 0,
+        "ports": [80, 443],
+        "service": {"type": "service",
+                    "service": {"srvce": "_http", "proto": "_tcp", "port": 80, "ttl": 60}},
+    }
+    o = start_orch(
+        ensemble3,
+        registration,
+        heartbeatInterval=100,
+        healthCheck={"command": "/bin/true", "interval": 100, "timeout": 1000,
+                     "threshold": 5, "period": 300000},
+    )
+    assert o.wait_registered(15000)
+    znodes = o.znodes()
+    svc_path = "/com/example/us-east/lb"
+    assert svc_path in znodes
+    svc = json.loads(ensemble3.get(svc_path)["data"])
+    assert svc["service"]["service"] == {"srvce": "_http", "proto": "_tcp", "ttl": 60, "port": 80}
+    host = json.loads(ensemble3.get(svc_path + "/lb0")["data"])
+    assert host["load_balancer"]["ports"] == [80, 443]
+    # the /bin/true gate keeps everything registered
+    import time as _t
+
+    _t.sleep(0.6)
+    assert all(ensemble3.get(n)["exists"] for n in znodes)
+    assert o.metrics()["unregisters"] == 0
+    assert o.metrics()["heartbeats"] >= 2
+    o.stop()
