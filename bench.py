@@ -150,7 +150,7 @@ def main():
         rc, err, znodes = prep.register_(client)
         if rc != 0:
             raise RuntimeError("rank %d register failed: %s" % (rank, err))
-        rc, rtt_us = client.heartbeat(znodes)
+        rc, rtt_us = prep.heartbeat(client)
         if rc != 0:
             raise RuntimeError("rank %d heartbeat failed: %s" % (rank, ra.error_name(rc)))
         return rtt_us

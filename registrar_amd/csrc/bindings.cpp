@@ -357,7 +357,7 @@ PYBIND11_MODULE(_core, m) {
       .def_property_readonly("host_payload",
                              [](const PreparedRegistration& p) { return py::bytes(p.host_payload); })
       .def("register_",
-           [](const PreparedRegistration& prep, zk::ZkClient& c, const std::string& log_level) {
+           [](PreparedRegistration& prep, zk::ZkClient& c, const std::string& log_level) {
              Logger log = make_logger("registrar", log_level);
              RegisterResult res;
              {
@@ -366,7 +366,19 @@ PYBIND11_MODULE(_core, m) {
              }
              return py::make_tuple(res.rc, res.error, res.znodes);
            },
-           py::arg("client"), py::arg("log_level") = "warn");
+           py::arg("client"), py::arg("log_level") = "warn")
+      .def("heartbeat",
+           [](PreparedRegistration& prep, zk::ZkClient& c, const py::dict& retry) {
+             zk::RetryPolicy rp = retry_from_dict(retry);
+             int64_t rtt = 0;
+             int rc;
+             {
+               py::gil_scoped_release rel;
+               rc = heartbeat_prepared(c, prep, rp, &rtt);
+             }
+             return py::make_tuple(rc, rtt);
+           },
+           py::arg("client"), py::arg("retry") = py::dict());
 
   m.def("register_node",
         [](zk::ZkClient& c, const std::string& registration_json, const std::string& log_level) {

@@ -236,10 +236,22 @@ PreparedRegistration prepare_registration(const RegistrationConfig& cfg) {
     op.flags = zk::kEphemeral;
     prep.wire_ops.push_back(std::move(op));
   }
+  prep.register_tpl = zk::ZkClient::make_template(prep.wire_ops);
+  prep.rest_tpl = zk::ZkClient::make_template(
+      std::vector<zk::ZkClient::MixedOp>(prep.wire_ops.begin() + static_cast<long>(prep.nodes.size()),
+                                         prep.wire_ops.end()));
+  prep.heartbeat_nodes = prep.nodes;
+  if (cfg.service) prep.heartbeat_nodes.push_back(prep.path);
+  prep.heartbeat_tpl = zk::ZkClient::make_exists_template(prep.heartbeat_nodes);
   return prep;
 }
 
-RegisterResult register_prepared(zk::ZkClient& client, const PreparedRegistration& prep, const Logger& log) {
+int heartbeat_prepared(zk::ZkClient& client, PreparedRegistration& prep, const zk::RetryPolicy& retry,
+                       int64_t* rtt_us) {
+  return client.heartbeat_template(prep.heartbeat_tpl, retry, rtt_us);
+}
+
+RegisterResult register_prepared(zk::ZkClient& client, PreparedRegistration& prep, const Logger& log) {
   RegisterResult result;
   const RegistrationConfig& cfg = prep.cfg;
   Logger rlog =
@@ -298,9 +310,7 @@ RegisterResult register_prepared(zk::ZkClient& client, const PreparedRegistratio
       return result;
     }
     std::this_thread::sleep_for(std::chrono::milliseconds(cfg.settle_ms));
-    std::vector<zk::ZkClient::MixedOp> rest(prep.wire_ops.begin() + static_cast<long>(n_nodes),
-                                            prep.wire_ops.end());
-    std::vector<int> rcs2 = client.submit_mixed(rest);
+    std::vector<int> rcs2 = client.submit_template(prep.rest_tpl);
     if (!check_dirs(rcs2, 0) || !check_creates(rcs2, n_dirs)) {
       rlog.debug("register: failed", {{"err", Json(result.error)}});
       return result;
@@ -309,7 +319,7 @@ RegisterResult register_prepared(zk::ZkClient& client, const PreparedRegistratio
     // no settle: the whole cleanup → mkdirp → create sequence is one
     // pipelined submission — ZooKeeper's per-session in-order processing
     // guarantees the same final state as the reference's staged barriers
-    std::vector<int> rcs = client.submit_mixed(prep.wire_ops);
+    std::vector<int> rcs = client.submit_template(prep.register_tpl);
     if (!check_cleanup(rcs, 0) || !check_dirs(rcs, n_nodes) || !check_creates(rcs, n_nodes + n_dirs)) {
       rlog.debug("register: failed", {{"err", Json(result.error)}});
       return result;
@@ -344,7 +354,8 @@ RegisterResult register_prepared(zk::ZkClient& client, const PreparedRegistratio
 }
 
 RegisterResult register_node(zk::ZkClient& client, const RegistrationConfig& cfg, const Logger& log) {
-  return register_prepared(client, prepare_registration(cfg), log);
+  PreparedRegistration prep = prepare_registration(cfg);
+  return register_prepared(client, prep, log);
 }
 
 int unregister_node(zk::ZkClient& client, const std::vector<std::string>& znodes, const Logger& log) {

@@ -105,12 +105,21 @@ struct PreparedRegistration {
   // creates) built once; with settle_ms == 0 it ships as a single pipelined
   // round trip (see zkclient submit_mixed)
   std::vector<zk::ZkClient::MixedOp> wire_ops;
+  // pre-serialized request streams (xid-patched per submission): the whole
+  // register sequence and the heartbeat exists-sweep
+  zk::ZkClient::BatchTemplate register_tpl;   // == wire_ops serialized
+  zk::ZkClient::BatchTemplate rest_tpl;       // wire_ops minus cleanup (settle path)
+  zk::ZkClient::BatchTemplate heartbeat_tpl;  // exists over heartbeat_nodes
+  std::vector<std::string> heartbeat_nodes;   // nodes (+ service path)
 };
 
 PreparedRegistration prepare_registration(const RegistrationConfig& cfg);
 
 // Full 5-step pipeline against a connected client.
-RegisterResult register_prepared(zk::ZkClient& client, const PreparedRegistration& prep, const Logger& log);
+RegisterResult register_prepared(zk::ZkClient& client, PreparedRegistration& prep, const Logger& log);
+// App-level heartbeat over the prepared node set (pre-serialized sweep).
+int heartbeat_prepared(zk::ZkClient& client, PreparedRegistration& prep,
+                       const zk::RetryPolicy& retry, int64_t* rtt_us);
 RegisterResult register_node(zk::ZkClient& client, const RegistrationConfig& cfg, const Logger& log);
 
 // Delete every znode; idempotent. Returns zk::kZOk or the first hard error.

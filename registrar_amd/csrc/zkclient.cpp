@@ -1000,6 +1000,121 @@ std::vector<int> ZkClient::submit_mixed(const std::vector<MixedOp>& ops) {
   return rcs;
 }
 
+ZkClient::BatchTemplate ZkClient::make_template(const std::vector<MixedOp>& mops) {
+  BatchTemplate t;
+  for (const auto& mo : mops) {
+    size_t start = t.buf.size();
+    t.buf.append(4, '\0');
+    t.xid_offsets.push_back(t.buf.size());
+    JuteWriter w(&t.buf);
+    w.write_int(0);  // xid placeholder
+    if (mo.op == kOpDelete) {
+      w.write_int(kOpDelete);
+      w.write_string(mo.path);
+      w.write_int(-1);
+    } else {
+      w.write_int(kOpCreate);
+      w.write_string(mo.path);
+      w.write_buffer(mo.data);
+      write_acl_vector(w, {ACL{}});
+      w.write_int(mo.flags);
+    }
+    uint32_t n = static_cast<uint32_t>(t.buf.size() - start - 4);
+    t.buf[start] = static_cast<char>(n >> 24);
+    t.buf[start + 1] = static_cast<char>(n >> 16);
+    t.buf[start + 2] = static_cast<char>(n >> 8);
+    t.buf[start + 3] = static_cast<char>(n);
+    t.ops.push_back(mo.op);
+  }
+  return t;
+}
+
+ZkClient::BatchTemplate ZkClient::make_exists_template(const std::vector<std::string>& paths) {
+  BatchTemplate t;
+  for (const auto& path : paths) {
+    size_t start = t.buf.size();
+    t.buf.append(4, '\0');
+    t.xid_offsets.push_back(t.buf.size());
+    JuteWriter w(&t.buf);
+    w.write_int(0);  // xid placeholder
+    w.write_int(kOpExists);
+    w.write_string(path);
+    w.write_bool(false);
+    uint32_t n = static_cast<uint32_t>(t.buf.size() - start - 4);
+    t.buf[start] = static_cast<char>(n >> 24);
+    t.buf[start + 1] = static_cast<char>(n >> 16);
+    t.buf[start + 2] = static_cast<char>(n >> 8);
+    t.buf[start + 3] = static_cast<char>(n);
+    t.ops.push_back(kOpExists);
+  }
+  return t;
+}
+
+std::vector<int> ZkClient::submit_template(BatchTemplate& t) {
+  size_t n = t.xid_offsets.size();
+  std::vector<int> rcs(n, kZConnectionLoss);
+  if (n == 0) return rcs;
+  BatchState st;
+  st.total = n;
+  st.rcs = &rcs;
+  BatchState* stp = &st;
+  impl_->loop.post([this, &t, stp, n] {
+    if (impl_->phase != Impl::Phase::Ready) {
+      for (size_t i = 0; i < n; i++)
+        stp->complete(i, impl_->phase == Impl::Phase::Stopped ? kZSessionExpired : kZConnectionLoss);
+      return;
+    }
+    // patch fresh xids in place, register pendings, append the whole stream
+    for (size_t i = 0; i < n; i++) {
+      int32_t xid = impl_->next_xid++;
+      size_t off = t.xid_offsets[i];
+      uint32_t u = static_cast<uint32_t>(xid);
+      t.buf[off] = static_cast<char>(u >> 24);
+      t.buf[off + 1] = static_cast<char>(u >> 16);
+      t.buf[off + 2] = static_cast<char>(u >> 8);
+      t.buf[off + 3] = static_cast<char>(u);
+      impl_->pending.push_back(
+          Impl::Pending{xid, t.ops[i], [stp, i](int rc, JuteReader*) { stp->complete(i, rc); }});
+    }
+    impl_->outbuf += t.buf;
+    impl_->schedule_flush();
+  });
+  st.wait();
+  return rcs;
+}
+
+int ZkClient::heartbeat_template(BatchTemplate& t, const RetryPolicy& retry, int64_t* rtt_us) {
+  Backoff bo;
+  bo.initial_ms = retry.initial_delay_ms;
+  bo.max_ms = retry.max_delay_ms;
+  bo.max_attempts = retry.max_attempts;
+  while (true) {
+    int64_t t0 = now_us();
+    std::vector<int> rcs = submit_template(t);
+    int rc = kZOk;
+    for (int r : rcs) {
+      if (r != kZOk) {
+        rc = r;
+        break;
+      }
+    }
+    if (rc == kZOk) {
+      if (rtt_us) *rtt_us = now_us() - t0;
+      return kZOk;
+    }
+    SessionState st = state();
+    if (rc == kZSessionExpired || st == SessionState::Expired || st == SessionState::Closed) return rc;
+    if (bo.exhausted()) return rc;
+    int64_t delay = bo.next_delay();
+    std::unique_lock<std::mutex> g(impl_->ev_mu);
+    impl_->ev_cv.wait_for(g, std::chrono::milliseconds(delay), [this] {
+      SessionState s2 = impl_->state.load();
+      return impl_->closed.load() || s2 == SessionState::Expired || s2 == SessionState::Closed;
+    });
+    if (impl_->closed.load()) return kZConnectionLoss;
+  }
+}
+
 int ZkClient::heartbeat(const std::vector<std::string>& nodes, const RetryPolicy& retry, int64_t* rtt_us) {
   Backoff bo;
   bo.initial_ms = retry.initial_delay_ms;

@@ -169,6 +169,23 @@ class ZkClient {
   };
   std::vector<int> submit_mixed(const std::vector<MixedOp>& ops);
 
+  // Pre-serialized batch: the full framed request stream built once, with
+  // per-request xid placeholders patched at submission time. For repeated
+  // identical batches (re-register cycles, heartbeats) the per-op cost
+  // drops to an xid patch + one buffer append.
+  struct BatchTemplate {
+    std::string buf;                  // framed requests, xids zeroed
+    std::vector<size_t> xid_offsets;  // offset of each request's xid field
+    std::vector<int32_t> ops;         // opcode per request (reply matching)
+  };
+  static BatchTemplate make_template(const std::vector<MixedOp>& ops);
+  static BatchTemplate make_exists_template(const std::vector<std::string>& paths);
+  // Patches fresh xids into `t` (mutated in place; one submission at a time
+  // per template) and pipelines the whole stream. Returns per-request rcs.
+  std::vector<int> submit_template(BatchTemplate& t);
+  // exists_many over a template with heartbeat-style bounded retry
+  int heartbeat_template(BatchTemplate& t, const RetryPolicy& retry, int64_t* rtt_us);
+
   // App-level heartbeat: parallel exists over `nodes` with bounded retry
   // (reference lib/zk.js:21-44: ≤5 attempts, 1 s → 30 s). Returns kZOk when a
   // round succeeded for every node; fills rtt_us with the successful round's
