@@ -315,19 +315,22 @@ struct ZkClient::Impl {
   // ---------------- frame processing ----------------
 
   bool read_frames() {
+    bool eof = false;
     char buf[65536];
     while (true) {
       ssize_t n = read(fd, buf, sizeof(buf));
       if (n > 0) {
         inbuf.append(buf, static_cast<size_t>(n));
       } else if (n == 0) {
-        on_connection_lost("eof");
-        return false;
+        // process buffered frames first: the server sends its final frames
+        // (e.g. an expired-handshake ConnectResponse) right before closing
+        eof = true;
+        break;
       } else {
         if (errno == EAGAIN || errno == EWOULDBLOCK) break;
         if (errno == EINTR) continue;
-        on_connection_lost("read error");
-        return false;
+        eof = true;
+        break;
       }
     }
     last_recv = now_ms();
@@ -350,6 +353,11 @@ struct ZkClient::Impl {
     if (inpos > 0 && fd >= 0) {
       inbuf.erase(0, inpos);
       inpos = 0;
+    }
+    if (eof) {
+      // a handler may already have torn the socket down (expiry path)
+      if (fd >= 0) on_connection_lost("eof");
+      return false;
     }
     return true;
   }
