@@ -171,10 +171,15 @@ struct Ensemble::Impl {
   }
 
   // Lock the shards of two paths without deadlock (index order; may be same).
+  // Exposes the shard refs so callers hash each path exactly once.
   struct TwoShardLock {
     std::unique_lock<std::mutex> a, b;
+    Shard* first;   // shard of p1
+    Shard* second;  // shard of p2
     TwoShardLock(Impl& impl, const std::string& p1, const std::string& p2) {
       size_t i1 = impl.shard_idx(p1), i2 = impl.shard_idx(p2);
+      first = &impl.shards[i1];
+      second = &impl.shards[i2];
       if (i1 == i2) {
         a = std::unique_lock<std::mutex>(impl.shards[i1].mu);
       } else {
@@ -540,7 +545,7 @@ struct Ensemble::Impl {
     } else {
       std::string parent = parent_path(req.path);
       TwoShardLock lk(*this, req.path, parent);
-      Shard& psh = shard_of(parent);
+      Shard& psh = *lk.second;
       auto pit = psh.nodes.find(parent);
       if (pit == psh.nodes.end()) {
         err = kZNoNode;
@@ -555,8 +560,8 @@ struct Ensemble::Impl {
         }
         // NB: with SEQUENCE the final path may hash to a different shard than
         // req.path; re-lock correctly in that (registrar-unused) corner
-        Shard& csh = shard_of(path);
-        bool same_lock = (&csh == &shard_of(req.path)) || (&csh == &psh);
+        Shard& csh = (req.flags & kSequence) ? shard_of(path) : *lk.first;
+        bool same_lock = (&csh == lk.first) || (&csh == &psh);
         std::unique_lock<std::mutex> extra;
         if (!same_lock) extra = std::unique_lock<std::mutex>(csh.mu, std::try_to_lock);
         if (!same_lock && !extra.owns_lock()) {
@@ -656,14 +661,14 @@ struct Ensemble::Impl {
     {
       std::string parent = parent_path(path);
       TwoShardLock lk(*this, path, parent);
-      Shard& csh = shard_of(path);
+      Shard& csh = *lk.first;
       auto it = csh.nodes.find(path);
       if (it == csh.nodes.end()) return false;
       if (!it->second.children.empty()) return false;  // re-check under lock
       int64_t z = zxid_counter.fetch_add(1, std::memory_order_relaxed) + 1;
       owner = it->second.stat.ephemeral_owner;
       csh.nodes.erase(it);
-      Shard& psh = shard_of(parent);
+      Shard& psh = *lk.second;
       auto pit = psh.nodes.find(parent);
       if (pit != psh.nodes.end()) {
         pit->second.children.erase(basename_of(path));
