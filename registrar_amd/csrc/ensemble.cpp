@@ -1217,11 +1217,9 @@ struct Ensemble::Impl {
     size_t lidx = idx % loops.size();
     std::promise<void> done;
     loops[lidx]->post([this, idx, lidx, &done] {
-      bool was_up = false;
       {
         std::lock_guard<std::mutex> g(admin_mu);
         if (idx < servers.size() && servers[idx].up) {
-          was_up = true;
           Server& s = servers[idx];
           loops[lidx]->del_fd(s.listen_fd);
           ::close(s.listen_fd);
