@@ -29,6 +29,10 @@
 
 using namespace registrar;
 
+#ifndef REGISTRAR_VERSION
+#define REGISTRAR_VERSION "0.1.0"
+#endif
+
 namespace {
 
 std::atomic<int> g_signal{0};
@@ -67,10 +71,14 @@ int main(int argc, char** argv) {
                                       {"verbose", no_argument, nullptr, 'v'},
                                       {"exit-on-expiry", no_argument, nullptr, 'e'},
                                       {"help", no_argument, nullptr, 'h'},
+                                      {"version", no_argument, nullptr, 'V'},
                                       {nullptr, 0, nullptr, 0}};
   int c;
-  while ((c = getopt_long(argc, argv, "f:veh", long_opts, nullptr)) != -1) {
+  while ((c = getopt_long(argc, argv, "f:vehV", long_opts, nullptr)) != -1) {
     switch (c) {
+      case 'V':
+        printf("registrard %s\n", REGISTRAR_VERSION);
+        return 0;
       case 'f':
         config_file = optarg;
         break;

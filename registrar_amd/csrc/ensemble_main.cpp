@@ -38,11 +38,13 @@ int main(int argc, char** argv) {
       cfg.latency_ms = atoi(argv[++i]);
     } else if (!strcmp(argv[i], "--election-ms") && i + 1 < argc) {
       cfg.election_ms = atoi(argv[++i]);
+    } else if (!strcmp(argv[i], "-b") && i + 1 < argc) {
+      cfg.bind_host = argv[++i];
     } else if (!strcmp(argv[i], "-v")) {
       cfg.log_level = LogLevel::Info;
     } else {
       fprintf(stderr,
-              "usage: %s [-n NSERVERS] [-p PORT]... [--tick-ms MS] "
+              "usage: %s [-n NSERVERS] [-p PORT]... [-b BINDHOST] [--tick-ms MS] "
               "[--latency-ms MS] [--election-ms MS] [-v]\n",
               argv[0]);
       return 1;
