@@ -287,7 +287,20 @@ void Orchestrator::heartbeat_loop() {
         client_.reset();
       }
       if (old) old->close();
-      if (!connect_and_register(false)) {
+      // bounded retries: a transient failure (storm chaos, mid-election)
+      // must not permanently kill registration (the reference gave up on
+      // any register error — lib/index.js:46-51)
+      bool recovered = false;
+      for (int attempt = 0; attempt < 5 && running_.load(); attempt++) {
+        if (connect_and_register(false)) {
+          recovered = true;
+          break;
+        }
+        std::unique_lock<std::mutex> g(wake_mu_);
+        wake_cv_.wait_for(g, std::chrono::milliseconds(500LL << attempt),
+                          [this] { return !running_.load(); });
+      }
+      if (!recovered) {
         if (running_.load()) expired_flag_.store(true);
         return;
       }

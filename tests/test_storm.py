@@ -154,3 +154,19 @@ def test_register_10k_znodes_scale(ensemble, client):
     assert rtt_us < 3_000_000  # still inside the heartbeat cadence envelope
     rcs = client.delete_many(znodes)
     assert all(r == ra.ZOK for r in rcs)
+
+
+def test_expiry_during_election_recovers(ensemble3):
+    """Worst-case storm: session expired AND leader killed at once; the
+    orchestrator's bounded re-register retries must ride out the chaos."""
+    cfg = orch_config(ensemble3, registration_1k(), heartbeatInterval=200)
+    o = ra.Orchestrator(json.dumps(cfg))
+    o.start()
+    assert o.wait_registered(30000)
+    sid1 = o.session_id()
+    ensemble3.kill_leader()
+    ensemble3.expire_session(sid1)
+    assert wait_for(lambda: ensemble3.ephemeral_count() == N
+                    and o.session_id() not in (0, sid1), timeout=40)
+    assert not o.expired()
+    o.stop()
