@@ -1,0 +1,159 @@
+"""Model-based testing: the ensemble + client against a pure-Python
+reference model of ZooKeeper's tree semantics (hypothesis stateful test).
+
+Covers create (±ephemeral), delete, setData, exists, getData, getChildren,
+with version checks, parent/child rules, and error codes — every operation's
+result is cross-checked against the model."""
+import json
+
+import pytest
+from hypothesis import HealthCheck, settings
+from hypothesis import strategies as st
+from hypothesis.stateful import RuleBasedStateMachine, initialize, invariant, rule
+
+import registrar_amd as ra
+
+# small path alphabet keeps collisions (the interesting cases) frequent
+NAMES = ["a", "b", "c", "d"]
+DATAS = [b"", b"x", b"payload-1", b'{"k":1}']
+
+
+def path_strategy():
+    return st.lists(st.sampled_from(NAMES), min_size=1, max_size=3).map(lambda p: "/" + "/".join(p))
+
+
+class Model:
+    """Pure-python ZK tree model: path -> (data, version, ephemeral)."""
+
+    def __init__(self):
+        self.nodes = {"/": [b"", 0, False]}
+
+    @staticmethod
+    def parent(path):
+        p = path.rsplit("/", 1)[0]
+        return p if p else "/"
+
+    def children(self, path):
+        if path not in self.nodes:
+            return None
+        base = path.rstrip("/")
+        out = []
+        for n in self.nodes:
+            if n == "/":
+                continue
+            pp = self.parent(n)
+            if pp == (base or "/"):
+                out.append(n.rsplit("/", 1)[1])
+        return sorted(out)
+
+    def create(self, path, data, ephemeral):
+        if path in self.nodes:
+            return ra.ZNODEEXISTS
+        par = self.parent(path)
+        if par not in self.nodes:
+            return ra.ZNONODE
+        if self.nodes[par][2]:
+            return ra.ZNOCHILDRENFOREPHEMERALS
+        self.nodes[path] = [data, 0, ephemeral]
+        return ra.ZOK
+
+    def delete(self, path, version):
+        if path not in self.nodes or path == "/":
+            return ra.ZNONODE
+        if self.children(path):
+            return ra.ZNOTEMPTY
+        if version != -1 and version != self.nodes[path][1]:
+            return ra.ZBADVERSION
+        del self.nodes[path]
+        return ra.ZOK
+
+    def set(self, path, data, version):
+        if path not in self.nodes:
+            return ra.ZNONODE
+        if version != -1 and version != self.nodes[path][1]:
+            return ra.ZBADVERSION
+        self.nodes[path][0] = data
+        self.nodes[path][1] += 1
+        return ra.ZOK
+
+
+class EnsembleMachine(RuleBasedStateMachine):
+    @initialize()
+    def setup(self):
+        self.ens = ra.Ensemble(servers=1, tick_ms=200)
+        self.ens.start()
+        host, port = self.ens.connect_string().rsplit(":", 1)
+        self.client = ra.ZkClient(servers=[(host, int(port))], session_timeout_ms=30000)
+        self.client.start()
+        assert self.client.wait_connected(15000)
+        self.model = Model()
+
+    def teardown(self):
+        if hasattr(self, "client"):
+            self.client.close()
+            self.ens.stop()
+
+    @rule(path=path_strategy(), data=st.sampled_from(DATAS), ephemeral=st.booleans())
+    def create(self, path, data, ephemeral):
+        rc, _ = self.client.create(path, data, ephemeral)
+        expect = self.model.create(path, data, ephemeral)
+        assert rc == expect, "create %s: got %s want %s" % (path, ra.error_name(rc), ra.error_name(expect))
+
+    @rule(path=path_strategy(), version=st.sampled_from([-1, 0, 1, 7]))
+    def delete(self, path, version):
+        rc = self.client.delete_(path, version)
+        expect = self.model.delete(path, version)
+        assert rc == expect, "delete %s v%d: got %s want %s" % (path, version, ra.error_name(rc),
+                                                               ra.error_name(expect))
+
+    @rule(path=path_strategy(), data=st.sampled_from(DATAS), version=st.sampled_from([-1, 0, 1, 7]))
+    def set_data(self, path, data, version):
+        rc = self.client.set(path, data, version)
+        expect = self.model.set(path, data, version)
+        assert rc == expect
+
+    @rule(path=path_strategy())
+    def check_exists(self, path):
+        rc, stat = self.client.exists(path)
+        if path in self.model.nodes:
+            assert rc == ra.ZOK
+            assert stat["version"] == self.model.nodes[path][1]
+            assert (stat["ephemeralOwner"] != 0) == self.model.nodes[path][2]
+        else:
+            assert rc == ra.ZNONODE
+
+    @rule(path=path_strategy())
+    def check_get(self, path):
+        rc, data, stat = self.client.get(path)
+        if path in self.model.nodes:
+            assert rc == ra.ZOK
+            assert data == self.model.nodes[path][0]
+            assert stat["dataLength"] == len(data)
+        else:
+            assert rc == ra.ZNONODE
+
+    @rule(path=path_strategy())
+    def check_children(self, path):
+        rc, ch = self.client.get_children(path)
+        expect = self.model.children(path)
+        if expect is None:
+            assert rc == ra.ZNONODE
+        else:
+            assert rc == ra.ZOK
+            assert ch == expect
+
+    @invariant()
+    def ephemeral_count_matches(self):
+        if not hasattr(self, "model"):
+            return
+        want = sum(1 for v in self.model.nodes.values() if v[2])
+        assert self.ens.ephemeral_count() == want
+
+
+TestEnsembleModel = EnsembleMachine.TestCase
+TestEnsembleModel.settings = settings(
+    max_examples=40,
+    stateful_step_count=40,
+    deadline=None,
+    suppress_health_check=[HealthCheck.too_slow],
+)
