@@ -95,7 +95,9 @@ def main():
         args.servers = 1 if world == 1 else (3 if world <= 4 else 5)
     ensemble = None
     if rank == 0:
-        ensemble = ra.Ensemble(servers=args.servers, tick_ms=100, max_session_timeout_ms=60000)
+        io_threads = int(os.environ.get("BENCH_IO_THREADS", "0"))
+        ensemble = ra.Ensemble(servers=args.servers, tick_ms=100, max_session_timeout_ms=60000,
+                               io_threads=io_threads)
         ensemble.start()
         connect = ensemble.connect_string()
         log("rank0 hosts ensemble at %s" % connect)

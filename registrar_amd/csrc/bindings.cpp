@@ -131,7 +131,7 @@ PYBIND11_MODULE(_core, m) {
   py::class_<zk::Ensemble>(m, "Ensemble")
       .def(py::init([](size_t servers, int tick_ms, int min_session_timeout_ms, int max_session_timeout_ms,
                        int latency_ms, int election_ms, const std::string& log_level,
-                       const std::vector<int>& ports, const std::string& bind_host) {
+                       const std::vector<int>& ports, const std::string& bind_host, int io_threads) {
              zk::EnsembleConfig cfg;
              if (!ports.empty()) {
                cfg.ports = ports;
@@ -139,6 +139,7 @@ PYBIND11_MODULE(_core, m) {
                cfg.ports.assign(servers, 0);
              }
              cfg.bind_host = bind_host;
+             cfg.io_threads = io_threads;
              cfg.tick_ms = tick_ms;
              cfg.min_session_timeout_ms = min_session_timeout_ms;
              cfg.max_session_timeout_ms = max_session_timeout_ms;
@@ -150,7 +151,7 @@ PYBIND11_MODULE(_core, m) {
            py::arg("servers") = 1, py::arg("tick_ms") = 100, py::arg("min_session_timeout_ms") = 400,
            py::arg("max_session_timeout_ms") = 60000, py::arg("latency_ms") = 0, py::arg("election_ms") = 0,
            py::arg("log_level") = "warn", py::arg("ports") = std::vector<int>{},
-           py::arg("bind_host") = "127.0.0.1")
+           py::arg("bind_host") = "127.0.0.1", py::arg("io_threads") = 0)
       .def("start", &zk::Ensemble::start, py::call_guard<py::gil_scoped_release>())
       .def("stop", &zk::Ensemble::stop, py::call_guard<py::gil_scoped_release>())
       .def("ports", &zk::Ensemble::ports)
