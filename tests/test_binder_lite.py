@@ -1,0 +1,125 @@
+"""binder_lite: DNS answers from registrar records — the discovery triangle
+end-to-end (registrar writes → binder-lite reads → DNS A/SRV answers)."""
+import json
+import socket
+import struct
+import time
+
+import pytest
+
+import registrar_amd as ra
+from registrar_amd.binder_lite import BinderLite, _encode_name
+from conftest import make_client, wait_for
+
+
+def dns_query(addr, name, qtype):
+    q = struct.pack(">HHHHHH", 0x1234, 0x0100, 1, 0, 0, 0)
+    q += _encode_name(name) + struct.pack(">HH", qtype, 1)
+    s = socket.socket(socket.AF_INET, socket.SOCK_DGRAM)
+    s.settimeout(5)
+    s.sendto(q, addr)
+    buf, _ = s.recvfrom(4096)
+    s.close()
+    txid, flags, qd, an, ns, ar = struct.unpack(">HHHHHH", buf[:12])
+    assert txid == 0x1234
+    return buf, flags & 0xF, an
+
+
+def parse_answers(buf, an):
+    # skip header + question
+    off = 12
+    while buf[off] != 0:
+        off += 1 + buf[off]
+    off += 1 + 4
+    answers = []
+    for _ in range(an):
+        off += 2  # name pointer
+        rtype, rclass, ttl, rdlen = struct.unpack(">HHIH", buf[off:off + 10])
+        off += 10
+        rdata = buf[off:off + rdlen]
+        off += rdlen
+        answers.append((rtype, ttl, rdata))
+    return answers
+
+
+@pytest.fixture
+def binder(ensemble):
+    servers = []
+    for hp in ensemble.connect_string().split(","):
+        host, port = hp.rsplit(":", 1)
+        servers.append((host, int(port)))
+    b = BinderLite(servers)
+    b.start()
+    yield b
+    b.stop()
+
+
+def register(client, domain, hosts, ttl=None, service=None):
+    for i, addr in enumerate(hosts):
+        reg = {"domain": domain, "type": "host", "adminIp": addr,
+               "hostname": "n%d" % i, "settleMs": 0}
+        if ttl is not None:
+            reg["ttl"] = ttl
+        if service is not None:
+            reg["service"] = service
+        rc, err, _ = ra.register_node(client, json.dumps(reg))
+        assert rc == ra.ZOK, err
+
+
+def test_a_records(ensemble, client, binder):
+    register(client, "web.dns.test", ["10.0.0.1", "10.0.0.2"], ttl=120)
+    buf, rcode, an = dns_query(binder.address, "web.dns.test", 1)
+    assert rcode == 0 and an == 2
+    answers = parse_answers(buf, an)
+    addrs = sorted(socket.inet_ntoa(rd) for _, _, rd in answers)
+    assert addrs == ["10.0.0.1", "10.0.0.2"]
+    assert all(ttl == 120 for _, ttl, _ in answers)
+
+
+def test_srv_records_and_ttl_precedence(ensemble, client, binder):
+    service = {"type": "service", "service": {"srvce": "_http", "proto": "_tcp", "port": 8080, "ttl": 45}}
+    register(client, "api.dns.test", ["10.1.0.1"], service=service)  # no host ttl
+    buf, rcode, an = dns_query(binder.address, "api.dns.test", 33)
+    assert rcode == 0 and an == 1
+    rtype, ttl, rdata = parse_answers(buf, an)[0]
+    assert rtype == 33
+    assert ttl == 45  # service-record ttl (host ttl absent)
+    prio, weight, port = struct.unpack(">HHH", rdata[:6])
+    assert port == 8080
+
+
+def test_host_ttl_wins_over_service(ensemble, client, binder):
+    service = {"type": "service", "service": {"srvce": "_http", "proto": "_tcp", "port": 80, "ttl": 45}}
+    register(client, "ttl.dns.test", ["10.2.0.1"], ttl=7, service=service)
+    buf, rcode, an = dns_query(binder.address, "ttl.dns.test", 1)
+    assert an == 1
+    assert parse_answers(buf, an)[0][1] == 7  # record-level ttl wins
+
+
+def test_nxdomain(binder):
+    _, rcode, an = dns_query(binder.address, "nosuch.dns.test", 1)
+    assert rcode == 3 and an == 0
+
+
+def test_liveness_via_ephemerals(ensemble, binder):
+    c2 = make_client(ensemble)
+    register(c2, "gone.dns.test", ["10.3.0.1"])
+    _, rcode, an = dns_query(binder.address, "gone.dns.test", 1)
+    assert an == 1
+    # instance dies ⇒ session closes ⇒ ephemeral vanishes ⇒ out of DNS
+    c2.close()
+    assert wait_for(lambda: ensemble.ephemeral_count() == 0, timeout=10)
+    _, rcode, an = dns_query(binder.address, "gone.dns.test", 1)
+    assert an == 0  # path (parents) persists: NOERROR, no answers
+
+
+def test_gpu_payload_does_not_break_answers(ensemble, client, binder):
+    reg = {"domain": "gpu.dns.test", "type": "host", "adminIp": "10.4.0.1", "hostname": "g0",
+           "settleMs": 0, "gpu": {"index": 0, "xgmiRank": 0, "uuid": "GPU-x"},
+           "service": {"type": "service", "service": {"srvce": "_infer", "proto": "_tcp", "port": 8000}}}
+    rc, err, _ = ra.register_node(client, json.dumps(reg))
+    assert rc == ra.ZOK
+    buf, rcode, an = dns_query(binder.address, "gpu.dns.test", 33)
+    assert an == 1
+    _, _, rdata = parse_answers(buf, an)[0]
+    assert struct.unpack(">HHH", rdata[:6])[2] == 8000
