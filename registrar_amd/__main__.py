@@ -9,6 +9,9 @@ Subcommands:
                                                synthetic ensemble or real ZK)
   gpus                                         show KFD GPU topology/xGMI ranks
   check    -f CONFIG                           validate a config file
+  binder   --servers HOST:PORT[,...] [--port N]  serve DNS A/SRV answers from
+                                               the registration tree
+                                               (binder_lite)
 """
 import argparse
 import json
@@ -125,6 +128,23 @@ def cmd_gpus(_args):
     return 0
 
 
+def cmd_binder(args):
+    from registrar_amd.binder_lite import BinderLite
+
+    b = BinderLite(_parse_servers(args.servers), host=args.host, port=args.port)
+    b.start()
+    print(json.dumps({"dns": "%s:%d" % b.address}), flush=True)
+    stop = []
+    signal.signal(signal.SIGINT, lambda *_: stop.append(1))
+    signal.signal(signal.SIGTERM, lambda *_: stop.append(1))
+    try:
+        while not stop:
+            time.sleep(0.1)
+    finally:
+        b.stop()
+    return 0
+
+
 def cmd_check(args):
     import registrar_amd as ra
 
@@ -166,6 +186,12 @@ def main(argv=None):
     ck = sub.add_parser("check", help="validate a config file")
     ck.add_argument("-f", "--file", required=True)
     ck.set_defaults(fn=cmd_check)
+
+    b = sub.add_parser("binder", help="serve DNS A/SRV answers from the registration tree")
+    b.add_argument("--servers", required=True, help="ZK host:port[,host:port...]")
+    b.add_argument("--host", default="127.0.0.1")
+    b.add_argument("--port", type=int, default=5353)
+    b.set_defaults(fn=cmd_binder)
 
     args = ap.parse_args(argv)
     return args.fn(args)
