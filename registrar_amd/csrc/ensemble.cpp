@@ -567,11 +567,12 @@ struct Ensemble::Impl {
         if (!same_lock && !extra.owns_lock()) {
           err = kZSystemError;  // pathological shard collision; not reachable
                                 // for non-sequence creates
-        } else if (csh.nodes.count(path)) {
+        } else if (auto [nit, inserted] = csh.nodes.try_emplace(path); !inserted) {
           err = kZNodeExists;
         } else {
+          // single hash+probe: the node was emplaced above; fill it in
           int64_t zz = zxid_counter.fetch_add(1, std::memory_order_relaxed) + 1;
-          ZNode n;
+          ZNode& n = nit->second;
           n.data = req.data;
           n.stat.czxid = zz;
           n.stat.mzxid = zz;
@@ -582,7 +583,6 @@ struct Ensemble::Impl {
             n.stat.ephemeral_owner = sid;
             made_ephemeral = true;
           }
-          csh.nodes[path] = std::move(n);
           ZNode& par = pit->second;
           par.children.insert(basename_of(path));
           par.stat.cversion++;

@@ -63,3 +63,24 @@ def test_daemon_subcommand(ensemble, tmp_path):
         out, _ = proc.communicate(timeout=15)
     assert proc.returncode == 0
     assert any('"event": "register"' in line for line in out.splitlines())
+
+
+def test_ensemble_subcommand():
+    proc = subprocess.Popen([sys.executable, "-m", "registrar_amd", "ensemble", "-n", "2"],
+                            stdout=subprocess.PIPE, text=True, env=ENV)
+    try:
+        line = proc.stdout.readline()
+        info = json.loads(line)
+        assert len(info["ports"]) == 2
+        import registrar_amd as ra
+
+        c = ra.ZkClient(servers=[("127.0.0.1", info["ports"][0])], connect_max_attempts=3)
+        c.start()
+        assert c.wait_connected(10000)
+        rc, _ = c.create("/via-cli", b"x")
+        assert rc == ra.ZOK
+        c.close()
+    finally:
+        proc.send_signal(signal.SIGTERM)
+        proc.wait(timeout=15)
+    assert proc.returncode == 0
