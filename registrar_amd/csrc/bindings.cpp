@@ -318,6 +318,34 @@ PYBIND11_MODULE(_core, m) {
              py::gil_scoped_release rel;
              return c.exists_many(paths, nullptr);
            })
+      .def("multi",
+           [](zk::ZkClient& c, const py::list& ops) {
+             std::vector<zk::ZkClient::MixedOp> mops;
+             for (auto item : ops) {
+               py::tuple t = item.cast<py::tuple>();
+               zk::ZkClient::MixedOp mo;
+               std::string kind = t[0].cast<std::string>();
+               mo.path = t[1].cast<std::string>();
+               if (kind == "delete") {
+                 mo.op = zk::kOpDelete;
+               } else if (kind == "create") {
+                 mo.op = zk::kOpCreate;
+                 if (t.size() > 2) mo.data = t[2].cast<std::string>();
+                 if (t.size() > 3 && t[3].cast<bool>()) mo.flags = zk::kEphemeral;
+               } else {
+                 throw std::runtime_error("multi op kind must be 'create' or 'delete'");
+               }
+               mops.push_back(std::move(mo));
+             }
+             std::vector<int> per_op;
+             int rc;
+             {
+               py::gil_scoped_release rel;
+               rc = c.multi(mops, &per_op);
+             }
+             return py::make_tuple(rc, per_op);
+           },
+           py::arg("ops"))
       .def("heartbeat",
            [](zk::ZkClient& c, const std::vector<std::string>& nodes, const py::dict& retry) {
              zk::RetryPolicy rp = retry_from_dict(retry);

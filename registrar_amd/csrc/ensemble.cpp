@@ -436,6 +436,9 @@ struct Ensemble::Impl {
         case kOpSetWatches:
           handle_set_watches(c, hdr.xid, r);
           break;
+        case kOpMulti:
+          handle_multi(c, hdr.xid, r);
+          break;
         case kOpCloseSession:
           handle_close_session(c, hdr.xid);
           break;
@@ -831,6 +834,326 @@ struct Ensemble::Impl {
       send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), err, nullptr);
   }
 
+  // multi (op 14): apply a create/delete/setData/check transaction
+  // atomically. All involved shards are locked in index order (a large multi
+  // effectively serializes — transactions are a correctness feature, not the
+  // throughput path), ops are validated against an overlay of the
+  // transaction's own effects, then applied; watches fire only on success.
+  struct MultiOpParsed {
+    int32_t type = 0;
+    std::string path;
+    std::string data;
+    int32_t flags = 0;
+    int32_t version = -1;
+  };
+
+  void handle_multi(Conn* c, int32_t xid, JuteReader& r) {
+    ops.create.fetch_add(0, std::memory_order_relaxed);  // (counted per sub-op below)
+    std::vector<MultiOpParsed> mops;
+    while (true) {
+      MultiHeader mh;
+      mh.deserialize(r);
+      if (mh.done) break;
+      MultiOpParsed mo;
+      mo.type = mh.type;
+      switch (mh.type) {
+        case kOpCreate: {
+          CreateRequest req;
+          req.deserialize(r);
+          mo.path = std::move(req.path);
+          mo.data = std::move(req.data);
+          mo.flags = req.flags;
+          break;
+        }
+        case kOpDelete: {
+          DeleteRequest req;
+          req.deserialize(r);
+          mo.path = std::move(req.path);
+          mo.version = req.version;
+          break;
+        }
+        case kOpSetData: {
+          SetDataRequest req;
+          req.deserialize(r);
+          mo.path = std::move(req.path);
+          mo.data = std::move(req.data);
+          mo.version = req.version;
+          break;
+        }
+        case 13: {  // check(path, version)
+          mo.path = r.read_string();
+          mo.version = r.read_int();
+          break;
+        }
+        default:
+          send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), kZMarshallingError, nullptr);
+          return;
+      }
+      mops.push_back(std::move(mo));
+    }
+
+    int64_t sid = sid_of(c);
+    std::vector<int32_t> results(mops.size(), kZOk);
+    std::vector<std::string> created_paths(mops.size());
+    std::vector<std::pair<std::string, int32_t>> data_events;   // path, event type
+    std::vector<std::string> child_events;                      // parent paths
+    std::vector<std::pair<std::string, bool>> eph_changes;      // path, added(true)/removed
+    int32_t txn_err = kZOk;
+
+    {
+      // lock every involved shard (paths + parents) in index order
+      std::set<size_t> idxs;
+      for (const auto& mo : mops) {
+        idxs.insert(shard_idx(mo.path));
+        idxs.insert(shard_idx(parent_path(mo.path)));
+      }
+      std::vector<std::unique_lock<std::mutex>> locks;
+      locks.reserve(idxs.size());
+      for (size_t i : idxs) locks.emplace_back(shards[i].mu);
+
+      // ---- validate against tree + overlay of txn effects ----
+      struct Overlay {
+        // path -> (exists, ephemeral); absent = defer to tree
+        std::unordered_map<std::string, std::pair<bool, bool>> state;
+        std::unordered_map<std::string, int> child_delta;
+      } ov;
+      auto tree_node = [&](const std::string& p) -> ZNode* {
+        auto& sh = shard_of(p);
+        auto it = sh.nodes.find(p);
+        return it == sh.nodes.end() ? nullptr : &it->second;
+      };
+      auto ov_exists = [&](const std::string& p) -> bool {
+        auto it = ov.state.find(p);
+        if (it != ov.state.end()) return it->second.first;
+        return tree_node(p) != nullptr;
+      };
+      auto ov_ephemeral = [&](const std::string& p) -> bool {
+        auto it = ov.state.find(p);
+        if (it != ov.state.end()) return it->second.second;
+        ZNode* n = tree_node(p);
+        return n && n->stat.ephemeral_owner != 0;
+      };
+      auto ov_child_count = [&](const std::string& p) -> int {
+        ZNode* n = tree_node(p);
+        int base = n ? static_cast<int>(n->children.size()) : 0;
+        auto it = ov.child_delta.find(p);
+        return base + (it == ov.child_delta.end() ? 0 : it->second);
+      };
+
+      for (size_t i = 0; i < mops.size(); i++) {
+        const auto& mo = mops[i];
+        int32_t err = kZOk;
+        if (!valid_path(mo.path) || mo.path == "/") {
+          err = (mo.type == kOpCreate && mo.path == "/") ? kZNodeExists : kZMarshallingError;
+        } else {
+          std::string parent = parent_path(mo.path);
+          switch (mo.type) {
+            case kOpCreate:
+              if (mo.flags & kSequence) {
+                err = kZMarshallingError;  // sequence-in-multi unsupported
+              } else if (!ov_exists(parent)) {
+                err = kZNoNode;
+              } else if (ov_ephemeral(parent)) {
+                err = kZNoChildrenForEphemerals;
+              } else if (ov_exists(mo.path)) {
+                err = kZNodeExists;
+              } else {
+                ov.state[mo.path] = {true, (mo.flags & kEphemeral) != 0};
+                ov.child_delta[parent]++;
+              }
+              break;
+            case kOpDelete: {
+              if (!ov_exists(mo.path)) {
+                err = kZNoNode;
+              } else if (ov_child_count(mo.path) > 0) {
+                err = kZNotEmpty;
+              } else {
+                bool created_in_txn = ov.state.count(mo.path) && ov.state[mo.path].first;
+                int32_t cur_version = created_in_txn ? 0 : (tree_node(mo.path) ? tree_node(mo.path)->stat.version : 0);
+                if (mo.version != -1 && mo.version != cur_version) {
+                  err = kZBadVersion;
+                } else {
+                  ov.state[mo.path] = {false, false};
+                  ov.child_delta[parent]--;
+                }
+              }
+              break;
+            }
+            case kOpSetData: {
+              if (!ov_exists(mo.path)) {
+                err = kZNoNode;
+              } else {
+                // version tracking through the overlay is approximate for
+                // txn-created nodes (version 0); real trees use the node's
+                ZNode* n = tree_node(mo.path);
+                int32_t cur = n ? n->stat.version : 0;
+                if (mo.version != -1 && mo.version != cur) err = kZBadVersion;
+              }
+              break;
+            }
+            case 13: {  // check
+              ZNode* n = tree_node(mo.path);
+              bool created_in_txn = ov.state.count(mo.path) && ov.state[mo.path].first;
+              if (!ov_exists(mo.path))
+                err = kZNoNode;
+              else if (mo.version != -1 && mo.version != (created_in_txn && !n ? 0 : (n ? n->stat.version : 0)))
+                err = kZBadVersion;
+              break;
+            }
+          }
+        }
+        if (err != kZOk && txn_err == kZOk) {
+          txn_err = err;
+          results[i] = err;
+        } else if (txn_err != kZOk) {
+          results[i] = kZRuntimeInconsistency;
+        }
+      }
+      if (txn_err != kZOk) {
+        // everything before the failure reports RuntimeInconsistency too
+        for (size_t i = 0; i < mops.size(); i++)
+          if (results[i] == kZOk) results[i] = kZRuntimeInconsistency;
+      } else {
+        // ---- apply (all locks still held) ----
+        for (size_t i = 0; i < mops.size(); i++) {
+          const auto& mo = mops[i];
+          std::string parent = parent_path(mo.path);
+          switch (mo.type) {
+            case kOpCreate: {
+              ops.create.fetch_add(1, std::memory_order_relaxed);
+              int64_t zz = zxid_counter.fetch_add(1, std::memory_order_relaxed) + 1;
+              Shard& csh = shard_of(mo.path);
+              ZNode& n = csh.nodes[mo.path];
+              n.data = mo.data;
+              n.stat = Stat{};
+              n.stat.czxid = zz;
+              n.stat.mzxid = zz;
+              n.stat.ctime = wall_ms();
+              n.stat.mtime = n.stat.ctime;
+              n.stat.data_length = static_cast<int32_t>(mo.data.size());
+              if (mo.flags & kEphemeral) {
+                n.stat.ephemeral_owner = sid;
+                eph_changes.push_back({mo.path, true});
+              }
+              Shard& psh = shard_of(parent);
+              auto pit = psh.nodes.find(parent);
+              if (pit != psh.nodes.end()) {
+                pit->second.children.insert(basename_of(mo.path));
+                pit->second.stat.cversion++;
+                pit->second.stat.pzxid = zz;
+                pit->second.stat.num_children = static_cast<int32_t>(pit->second.children.size());
+              }
+              data_events.push_back({mo.path, kEventNodeCreated});
+              child_events.push_back(parent);
+              created_paths[i] = mo.path;
+              break;
+            }
+            case kOpDelete: {
+              ops.del.fetch_add(1, std::memory_order_relaxed);
+              int64_t zz = zxid_counter.fetch_add(1, std::memory_order_relaxed) + 1;
+              Shard& csh = shard_of(mo.path);
+              auto it = csh.nodes.find(mo.path);
+              if (it != csh.nodes.end()) {
+                if (it->second.stat.ephemeral_owner != 0) eph_changes.push_back({mo.path, false});
+                csh.nodes.erase(it);
+              }
+              Shard& psh = shard_of(parent);
+              auto pit = psh.nodes.find(parent);
+              if (pit != psh.nodes.end()) {
+                pit->second.children.erase(basename_of(mo.path));
+                pit->second.stat.cversion++;
+                pit->second.stat.pzxid = zz;
+                pit->second.stat.num_children = static_cast<int32_t>(pit->second.children.size());
+              }
+              data_events.push_back({mo.path, kEventNodeDeleted});
+              child_events.push_back(parent);
+              break;
+            }
+            case kOpSetData: {
+              ops.set_data.fetch_add(1, std::memory_order_relaxed);
+              int64_t zz = zxid_counter.fetch_add(1, std::memory_order_relaxed) + 1;
+              Shard& csh = shard_of(mo.path);
+              auto it = csh.nodes.find(mo.path);
+              if (it != csh.nodes.end()) {
+                it->second.data = mo.data;
+                it->second.stat.mzxid = zz;
+                it->second.stat.mtime = wall_ms();
+                it->second.stat.version++;
+                it->second.stat.data_length = static_cast<int32_t>(mo.data.size());
+              }
+              data_events.push_back({mo.path, kEventNodeDataChanged});
+              break;
+            }
+            default:
+              break;  // check: no effect
+          }
+        }
+        // fire watches while holding the shard locks (same discipline as
+        // the single-op handlers: shard → session_mu → conns_mu → out_mu)
+        for (auto& [path, ev] : data_events) fire_data_watches_locked(shard_of(path), path, ev);
+        std::set<std::string> fired;
+        for (auto& parent : child_events)
+          if (fired.insert(parent).second) fire_child_watches_locked(shard_of(parent), parent);
+      }
+    }
+
+    // session ephemeral bookkeeping outside the shard locks (same alive-flag
+    // handshake as single-op create)
+    if (txn_err == kZOk && !eph_changes.empty()) {
+      SessionPtr s = c->session;
+      if (s) {
+        bool ok;
+        {
+          std::lock_guard<std::mutex> eg(s->eph_mu);
+          for (auto& [path, added] : eph_changes) {
+            if (added)
+              s->ephemerals.insert(path);
+            else
+              s->ephemerals.erase(path);
+          }
+        }
+        ok = s->alive.load(std::memory_order_acquire);
+        if (!ok) {
+          // session died mid-txn: roll the created ephemerals back
+          std::lock_guard<std::mutex> eg(s->eph_mu);
+          for (auto& [path, added] : eph_changes)
+            if (added && s->ephemerals.erase(path)) delete_node(path, s);
+        }
+      }
+    }
+
+    int64_t z = zxid_counter.load(std::memory_order_relaxed);
+    send_reply_with_body(c, xid, z, txn_err, [&](JuteWriter& w) {
+      for (size_t i = 0; i < mops.size(); i++) {
+        MultiHeader mh;
+        mh.done = false;
+        if (txn_err != kZOk) {
+          mh.type = -1;
+          mh.err = results[i];
+          mh.serialize(w);
+          w.write_int(results[i]);  // ErrorResult
+        } else {
+          mh.type = mops[i].type;
+          mh.err = 0;
+          mh.serialize(w);
+          if (mops[i].type == kOpCreate) w.write_string(created_paths[i]);
+          if (mops[i].type == kOpSetData) {
+            Shard& sh = shard_of(mops[i].path);
+            auto it = sh.nodes.find(mops[i].path);
+            Stat st;
+            if (it != sh.nodes.end()) st = it->second.stat;
+            st.serialize(w);
+          }
+        }
+      }
+      MultiHeader end;
+      end.type = -1;
+      end.done = true;
+      end.err = -1;
+      end.serialize(w);
+    });
+  }
+
   // setWatches: re-arm a reconnected session's watches, firing synthetic
   // events for anything that changed past relative_zxid while it was away
   void handle_set_watches(Conn* c, int32_t xid, JuteReader& r) {
@@ -1049,6 +1372,22 @@ struct Ensemble::Impl {
 
   void send_reply(Conn* c, int32_t xid, int64_t zxid, int32_t err, std::nullptr_t) {
     send_reply<std::nullptr_t>(c, xid, zxid, err, nullptr);
+  }
+
+  // multi replies carry a body even on failure (per-op error results)
+  template <typename BodyFn>
+  void send_reply_with_body(Conn* c, int32_t xid, int64_t zxid, int32_t err, BodyFn body) {
+    std::string pkt;
+    begin_packet(&pkt);
+    JuteWriter w(&pkt);
+    ReplyHeader hdr;
+    hdr.xid = xid;
+    hdr.zxid = zxid;
+    hdr.err = err;
+    hdr.serialize(w);
+    body(w);
+    frame_packet(&pkt);
+    send_packet(c, std::move(pkt));
   }
 
   // Any thread. Latency injection routes through the owner loop's timers

@@ -519,6 +519,28 @@ struct GetChildrenResponse {
   }
 };
 
+// multi (op 14): a transaction of create/delete/setData/check ops applied
+// atomically. Wire format: repeated {MultiHeader{type, done=false, err} +
+// op record}, terminated by MultiHeader{-1, true, -1}. The response uses
+// the same shape with result records (create → path; delete/check →
+// nothing; error result type -1 → int error code).
+struct MultiHeader {
+  int32_t type = -1;
+  bool done = true;
+  int32_t err = -1;
+
+  void serialize(JuteWriter& w) const {
+    w.write_int(type);
+    w.write_bool(done);
+    w.write_int(err);
+  }
+  void deserialize(JuteReader& r) {
+    type = r.read_int();
+    done = r.read_bool();
+    err = r.read_int();
+  }
+};
+
 // setWatches (op 101, xid -8): re-arm watches after a same-session
 // reconnect; the server fires synthetic events for changes that happened
 // after relative_zxid while the client was disconnected.

@@ -128,6 +128,7 @@ RegistrationConfig parse_registration(const Json& j) {
   if (const Json* settle = j.find("settleMs")) {
     if (settle->is_number()) cfg.settle_ms = settle->as_int();
   }
+  cfg.atomic_swap = j.get_bool("atomicSwap", false);
   return cfg;
 }
 
@@ -300,7 +301,57 @@ RegisterResult register_prepared(zk::ZkClient& client, PreparedRegistration& pre
     return true;
   };
 
-  if (cfg.settle_ms > 0) {
+  if (cfg.atomic_swap) {
+    // ---- atomic swap (beyond the reference) ----
+    // 1) parents (idempotent, persistent, usually a no-op)
+    {
+      std::vector<std::string> datas(prep.dirs.size());
+      std::vector<int> rcs = client.create_many(prep.dirs, datas, 0);
+      if (!check_dirs(rcs, 0)) {
+        rlog.debug("setupDirectories: failed", {{"err", Json(result.error)}});
+        return result;
+      }
+    }
+    // 2) read current state, 3) swap in one transaction; retry on races
+    for (int attempt = 0;; attempt++) {
+      std::vector<int> ex = client.exists_many(prep.nodes, nullptr);
+      std::vector<zk::ZkClient::MixedOp> ops;
+      ops.reserve(prep.nodes.size() * 2);
+      for (size_t i = 0; i < prep.nodes.size(); i++) {
+        if (ex[i] == zk::kZOk) {
+          zk::ZkClient::MixedOp del;
+          del.op = zk::kOpDelete;
+          del.path = prep.nodes[i];
+          ops.push_back(std::move(del));
+        } else if (ex[i] != zk::kZNoNode) {
+          result.rc = ex[i];
+          result.error = std::string("atomicSwap: exists ") + prep.nodes[i] + " failed: " +
+                         zk::error_name(ex[i]);
+          return result;
+        }
+      }
+      for (size_t i = 0; i < prep.nodes.size(); i++) {
+        zk::ZkClient::MixedOp cr;
+        cr.op = zk::kOpCreate;
+        cr.path = prep.nodes[i];
+        cr.data = prep.host_payload;
+        cr.flags = zk::kEphemeral;
+        ops.push_back(std::move(cr));
+      }
+      std::vector<int> per_op;
+      int rc = client.multi(ops, &per_op);
+      if (rc == zk::kZOk) break;
+      bool race = (rc == zk::kZNodeExists || rc == zk::kZNoNode);
+      if (!race || attempt >= 3) {
+        result.rc = rc;
+        result.error = std::string("atomicSwap: multi failed: ") + zk::error_name(rc);
+        rlog.debug("atomicSwap: failed", {{"err", Json(result.error)}});
+        return result;
+      }
+      // another session created/removed one of our nodes between the read
+      // and the transaction — re-read and retry
+    }
+  } else if (cfg.settle_ms > 0) {
     // settle configured: a real barrier after cleanup (reference semantics,
     // fixed 1000 ms at lib/register.js:232-235), then the remaining stages
     // in one pipelined round trip

@@ -60,6 +60,11 @@ struct RegistrationConfig {
   int64_t settle_ms = 1000;            // post-cleanup watcher settle delay
                                        // (fixed 1000 ms in the reference,
                                        // lib/register.js:232-235)
+  // atomic swap (beyond the reference): replace stale znodes and create the
+  // new set in ONE ZooKeeper multi transaction — consumers never observe a
+  // partially-registered domain (the reference's cleanup→create gap is what
+  // its 1 s settle delay papered over). settle_ms is moot in this mode.
+  bool atomic_swap = false;
 };
 
 // Parse + validate a `registration` JSON block (schema: SURVEY.md §2.5).

@@ -451,10 +451,9 @@ struct ZkClient::Impl {
         return false;
       }
       if (p.done) {
-        if (hdr.err == kZOk)
-          p.done(kZOk, &r);
-        else
-          p.done(hdr.err, nullptr);
+        // the reader is passed even on error replies: multi responses carry
+        // per-op error results in the body (callbacks gate on rc themselves)
+        p.done(hdr.err, &r);
       }
       return true;
     } catch (const std::exception& e) {
@@ -1006,6 +1005,62 @@ std::vector<int> ZkClient::submit_mixed(const std::vector<MixedOp>& ops) {
   });
   st.wait();
   return rcs;
+}
+
+int ZkClient::multi(const std::vector<MixedOp>& mops, std::vector<int>* per_op) {
+  if (per_op) per_op->assign(mops.size(), kZConnectionLoss);
+  std::promise<int> done;
+  impl_->loop.post([this, &mops, per_op, &done] {
+    impl_->submit_op(
+        kOpMulti,
+        [&mops](JuteWriter& w) {
+          for (const auto& mo : mops) {
+            MultiHeader mh;
+            mh.type = mo.op;
+            mh.done = false;
+            mh.err = -1;
+            mh.serialize(w);
+            if (mo.op == kOpDelete) {
+              w.write_string(mo.path);
+              w.write_int(-1);
+            } else {
+              w.write_string(mo.path);
+              w.write_buffer(mo.data);
+              write_acl_vector(w, {ACL{}});
+              w.write_int(mo.flags);
+            }
+          }
+          MultiHeader end;
+          end.serialize(w);
+        },
+        [per_op, &done, n = mops.size()](int rc, JuteReader* r) {
+          // parse per-op results when a body is present (success or txn abort)
+          if (r) {
+            size_t i = 0;
+            try {
+              while (r->remaining() > 0 && i < n) {
+                MultiHeader mh;
+                mh.deserialize(*r);
+                if (mh.done) break;
+                int op_rc = kZOk;
+                if (mh.type == -1) {
+                  op_rc = r->read_int();  // ErrorResult
+                } else if (mh.type == kOpCreate) {
+                  r->read_string();
+                } else if (mh.type == kOpSetData) {
+                  Stat st;
+                  st.deserialize(*r);
+                }
+                if (per_op && i < per_op->size()) (*per_op)[i] = op_rc;
+                i++;
+              }
+            } catch (const std::exception&) {
+            }
+          }
+          done.set_value(rc);
+        });
+  });
+  return done.get_future().get();
 }
 
 ZkClient::BatchTemplate ZkClient::make_template(const std::vector<MixedOp>& mops) {

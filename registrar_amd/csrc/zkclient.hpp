@@ -169,6 +169,12 @@ class ZkClient {
   };
   std::vector<int> submit_mixed(const std::vector<MixedOp>& ops);
 
+  // ZooKeeper multi (op 14): apply the create/delete sequence as ONE atomic
+  // transaction. Returns the transaction rc (kZOk = all applied) and fills
+  // per_op with each op's result (the failing op's error; others
+  // RuntimeInconsistency on abort).
+  int multi(const std::vector<MixedOp>& ops, std::vector<int>* per_op = nullptr);
+
   // Pre-serialized batch: the full framed request stream built once, with
   // per-request xid placeholders patched at submission time. For repeated
   // identical batches (re-register cycles, heartbeats) the per-op cost
