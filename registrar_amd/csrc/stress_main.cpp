@@ -79,6 +79,21 @@ int main(int argc, char** argv) {
         client.exists(res.znodes[0], nullptr, true);
         std::vector<std::string> ch;
         client.get_children(domain_to_path(reg.domain), &ch, true);
+        // exercise multi under chaos too: atomic delete+create of one node
+        {
+          std::vector<zk::ZkClient::MixedOp> mops;
+          zk::ZkClient::MixedOp d;
+          d.op = zk::kOpDelete;
+          d.path = res.znodes[0];
+          mops.push_back(d);
+          zk::ZkClient::MixedOp cr;
+          cr.op = zk::kOpCreate;
+          cr.path = res.znodes[0];
+          cr.data = "swap";
+          cr.flags = zk::kEphemeral;
+          mops.push_back(cr);
+          client.multi(mops, nullptr);  // rc may be conn-loss under chaos
+        }
         zk::RetryPolicy rp;
         rp.max_attempts = 1;
         rp.initial_delay_ms = 10;

@@ -112,6 +112,27 @@ class EnsembleMachine(RuleBasedStateMachine):
         expect = self.model.set(path, data, version)
         assert rc == expect
 
+    @rule(ops=st.lists(
+        st.tuples(st.sampled_from(["create", "delete"]), st.lists(st.sampled_from(NAMES), min_size=1,
+                  max_size=3).map(lambda p: "/" + "/".join(p)), st.sampled_from(DATAS), st.booleans()),
+        min_size=1, max_size=4))
+    def multi(self, ops):
+        mops = [(kind, path, data, eph) for kind, path, data, eph in ops]
+        rc, per_op = self.client.multi(mops)
+        # model: simulate on a copy; atomic = all-or-nothing
+        import copy
+
+        trial = copy.deepcopy(self.model)
+        expect = ra.ZOK
+        for kind, path, data, eph in mops:
+            r = trial.create(path, data, eph) if kind == "create" else trial.delete(path, -1)
+            if r != ra.ZOK:
+                expect = r
+                break
+        assert rc == expect, "multi: got %s want %s (%r)" % (ra.error_name(rc), ra.error_name(expect), mops)
+        if rc == ra.ZOK:
+            self.model = trial
+
     @rule(path=path_strategy())
     def check_exists(self, path):
         rc, stat = self.client.exists(path)
