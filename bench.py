@@ -193,6 +193,8 @@ def main():
         p99_ms = percentile(all_rtts, 99) / 1000.0
         result = {
             "metric": "ephemeral registrations/sec",
+            "baseline_metric": "ephemeral registrations/sec + p50 ZK heartbeat RTT, "
+                               "1k znodes, 1/2/4/8 procs",  # BASELINE.json's name
             "value": round(value, 1),
             "unit": "registrations/s",
             "n_gpus": world,
