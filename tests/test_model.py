@@ -82,11 +82,14 @@ class EnsembleMachine(RuleBasedStateMachine):
     def setup(self):
         self.ens = ra.Ensemble(servers=1, tick_ms=200)
         self.ens.start()
+        self.model = Model()
+        self._new_client()
+
+    def _new_client(self):
         host, port = self.ens.connect_string().rsplit(":", 1)
         self.client = ra.ZkClient(servers=[(host, int(port))], session_timeout_ms=30000)
         self.client.start()
         assert self.client.wait_connected(15000)
-        self.model = Model()
 
     def teardown(self):
         if hasattr(self, "client"):
@@ -132,6 +135,23 @@ class EnsembleMachine(RuleBasedStateMachine):
         assert rc == expect, "multi: got %s want %s (%r)" % (ra.error_name(rc), ra.error_name(expect), mops)
         if rc == ra.ZOK:
             self.model = trial
+
+    @rule()
+    def expire_session(self):
+        # session death: every ephemeral this session owns vanishes (all
+        # ephemerals in this machine are ours — single client); the client is
+        # terminal and must be replaced, exactly like a daemon would
+        import time
+
+        self.ens.expire_session(self.client.session_id())
+        for p in [p for p, v in self.model.nodes.items() if v[2]]:
+            del self.model.nodes[p]
+        deadline = time.monotonic() + 10
+        while time.monotonic() < deadline and self.client.state() != "expired":
+            time.sleep(0.02)
+        assert self.client.state() == "expired"
+        self.client.close()
+        self._new_client()
 
     @rule(path=path_strategy())
     def check_exists(self, path):

@@ -30,6 +30,7 @@ def main():
     ap.add_argument("--seconds", type=int, default=60)
     ap.add_argument("--znodes", type=int, default=100)
     ap.add_argument("--gpu", action="store_true", help="use the gpu-liveness health gate")
+    ap.add_argument("--atomic", action="store_true", help="atomicSwap registration mode")
     args = ap.parse_args()
 
     ens = ra.Ensemble(servers=3, tick_ms=100, election_ms=200, min_session_timeout_ms=1000)
@@ -48,6 +49,7 @@ def main():
             "hostname": "e0",
             "settleMs": 0,
             "aliases": ["a%03d.%s" % (i, domain) for i in range(args.znodes - 1)],
+            "atomicSwap": bool(args.atomic),
         },
         "zookeeper": {"servers": servers, "timeout": 4000, "connectTimeout": 1000},
         "heartbeatInterval": 500,
@@ -135,6 +137,7 @@ def main():
             "seconds": args.seconds,
             "znodes": args.znodes,
             "gpu_gate": args.gpu,
+            "atomic_swap": bool(args.atomic),
             "actions": actions,
             "failures": failures,
             "daemon_exit": proc.returncode,
