@@ -7,6 +7,7 @@
 // which is the only thread-safe entry point.
 #pragma once
 
+#include <stdio.h>
 #include <sys/epoll.h>
 #include <sys/eventfd.h>
 #include <time.h>
@@ -162,7 +163,7 @@ class EventLoop {
           auto it = fd_cbs_.find(fd);
           if (it != fd_cbs_.end()) cb = it->second;  // copy: cb may del_fd(fd)
         }
-        if (cb) cb(events[i].events);
+        if (cb) guarded([&] { cb(events[i].events); });
       }
       fire_timers();
       if (n == static_cast<int>(events.size())) events.resize(events.size() * 2);
@@ -181,13 +182,27 @@ class EventLoop {
     (void)r;
   }
 
+  // A throwing callback must not take down the loop thread (and with it the
+  // daemon): log to stderr and keep serving. Logic errors still surface in
+  // tests via the message.
+  template <typename Fn>
+  void guarded(Fn&& fn) {
+    try {
+      fn();
+    } catch (const std::exception& e) {
+      fprintf(stderr, "[eventloop] callback threw: %s\n", e.what());
+    } catch (...) {
+      fprintf(stderr, "[eventloop] callback threw (non-std exception)\n");
+    }
+  }
+
   void drain_posted() {
     std::deque<std::function<void()>> q;
     {
       std::lock_guard<std::mutex> g(post_mu_);
       q.swap(posted_);
     }
-    for (auto& fn : q) fn();
+    for (auto& fn : q) guarded(fn);
   }
 
   int next_timeout_ms() {
@@ -214,7 +229,7 @@ class EventLoop {
       if (it == timers_.end()) continue;
       auto fn = std::move(it->second);
       timers_.erase(it);
-      fn();
+      guarded(fn);
     }
   }
 
