@@ -147,12 +147,18 @@ def main():
     # pipeline, exactly like the daemon's re-register path
     prep = ra.PreparedRegistration(json.dumps(registration))
 
+    phase_time = {"reg": 0.0, "hb": 0.0}
+
     def step():
         """One full re-register of all znodes + one 1k-node heartbeat."""
+        t0 = time.perf_counter()
         rc, err, znodes = prep.register_(client)
+        t1 = time.perf_counter()
         if rc != 0:
             raise RuntimeError("rank %d register failed: %s" % (rank, err))
         rc, rtt_us = prep.heartbeat(client)
+        phase_time["reg"] += t1 - t0
+        phase_time["hb"] += time.perf_counter() - t1
         if rc != 0:
             raise RuntimeError("rank %d heartbeat failed: %s" % (rank, ra.error_name(rc)))
         return rtt_us
@@ -179,9 +185,10 @@ def main():
         all_rtts = [r for lst in gathered for r in lst]
         if os.environ.get("BENCH_DEBUG"):
             per = [None] * world
-            dist.all_gather_object(per, elapsed)
+            dist.all_gather_object(per, (elapsed, phase_time["reg"], phase_time["hb"]))
             if rank == 0:
-                log("per-rank elapsed: %s" % ", ".join("%.3f" % e for e in per))
+                log("per-rank (elapsed, reg, hb): %s"
+                    % ", ".join("(%.3f, %.3f, %.3f)" % e for e in per))
     else:
         elapsed_max = elapsed
         all_rtts = rtts_us
