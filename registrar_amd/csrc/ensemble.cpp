@@ -848,7 +848,6 @@ struct Ensemble::Impl {
   };
 
   void handle_multi(Conn* c, int32_t xid, JuteReader& r) {
-    ops.create.fetch_add(0, std::memory_order_relaxed);  // (counted per sub-op below)
     std::vector<MultiOpParsed> mops;
     while (true) {
       MultiHeader mh;
@@ -897,7 +896,9 @@ struct Ensemble::Impl {
     std::vector<std::string> created_paths(mops.size());
     std::vector<std::pair<std::string, int32_t>> data_events;   // path, event type
     std::vector<std::string> child_events;                      // parent paths
-    std::vector<std::pair<std::string, bool>> eph_changes;      // path, added(true)/removed
+    // ephemeral bookkeeping: (path, added, owner-session) — a delete may
+    // remove ANOTHER session's ephemeral; its owner's set must be updated
+    std::vector<std::tuple<std::string, bool, int64_t>> eph_changes;
     int32_t txn_err = kZOk;
 
     {
@@ -1033,7 +1034,7 @@ struct Ensemble::Impl {
               n.stat.data_length = static_cast<int32_t>(mo.data.size());
               if (mo.flags & kEphemeral) {
                 n.stat.ephemeral_owner = sid;
-                eph_changes.push_back({mo.path, true});
+                eph_changes.push_back({mo.path, true, sid});
               }
               Shard& psh = shard_of(parent);
               auto pit = psh.nodes.find(parent);
@@ -1054,7 +1055,8 @@ struct Ensemble::Impl {
               Shard& csh = shard_of(mo.path);
               auto it = csh.nodes.find(mo.path);
               if (it != csh.nodes.end()) {
-                if (it->second.stat.ephemeral_owner != 0) eph_changes.push_back({mo.path, false});
+                if (it->second.stat.ephemeral_owner != 0)
+                  eph_changes.push_back({mo.path, false, it->second.stat.ephemeral_owner});
                 csh.nodes.erase(it);
               }
               Shard& psh = shard_of(parent);
@@ -1098,27 +1100,30 @@ struct Ensemble::Impl {
     }
 
     // session ephemeral bookkeeping outside the shard locks (same alive-flag
-    // handshake as single-op create)
+    // handshake as single-op create); deletes may belong to OTHER sessions
     if (txn_err == kZOk && !eph_changes.empty()) {
-      SessionPtr s = c->session;
-      if (s) {
-        bool ok;
-        {
-          std::lock_guard<std::mutex> eg(s->eph_mu);
-          for (auto& [path, added] : eph_changes) {
-            if (added)
-              s->ephemerals.insert(path);
-            else
-              s->ephemerals.erase(path);
-          }
-        }
-        ok = s->alive.load(std::memory_order_acquire);
-        if (!ok) {
-          // session died mid-txn: roll the created ephemerals back
-          std::lock_guard<std::mutex> eg(s->eph_mu);
-          for (auto& [path, added] : eph_changes)
-            if (added && s->ephemerals.erase(path)) delete_node(path, s);
-        }
+      SessionPtr mine = c->session;
+      auto owner_session = [&](int64_t owner) -> SessionPtr {
+        if (mine && mine->id == owner) return mine;
+        std::lock_guard<std::mutex> g(session_mu);
+        auto sit = sessions.find(owner);
+        return sit == sessions.end() ? nullptr : sit->second;
+      };
+      for (auto& [path, added, owner] : eph_changes) {
+        SessionPtr s = owner_session(owner);
+        if (!s) continue;  // owner already dead; its drain saw (or will
+                           // no-op on) the node
+        std::lock_guard<std::mutex> eg(s->eph_mu);
+        if (added)
+          s->ephemerals.insert(path);
+        else
+          s->ephemerals.erase(path);
+      }
+      if (mine && !mine->alive.load(std::memory_order_acquire)) {
+        // our session died mid-txn: roll our created ephemerals back
+        std::lock_guard<std::mutex> eg(mine->eph_mu);
+        for (auto& [path, added, owner] : eph_changes)
+          if (added && owner == mine->id && mine->ephemerals.erase(path)) delete_node(path, mine);
       }
     }
 

@@ -186,3 +186,22 @@ def test_atomic_swap_via_orchestrator(ensemble):
     assert wait_for(lambda: ensemble.ephemeral_count() == 20
                     and o.session_id() not in (0, sid1), timeout=20)
     o.stop()
+
+
+def test_multi_delete_of_other_sessions_ephemeral(ensemble):
+    """Deleting another session's ephemeral in a txn must update that owner's
+    bookkeeping: when the owner later dies, counts stay exact."""
+    owner = make_client(ensemble)
+    actor = make_client(ensemble)
+    owner.mkdirp("/x")
+    owner.create("/x/theirs", b"", True)
+    assert ensemble.ephemeral_count() == 1
+    rc, _ = actor.multi([("delete", "/x/theirs"), ("create", "/x/mine", b"", True)])
+    assert rc == ra.ZOK
+    assert ensemble.ephemeral_count() == 1  # theirs removed from the owner's books
+    ensemble.expire_session(owner.session_id())
+    assert wait_for(lambda: owner.state() == "expired", timeout=10)
+    assert ensemble.get("/x/mine")["exists"]  # actor's node untouched by owner's death
+    assert ensemble.ephemeral_count() == 1
+    actor.close()
+    owner.close()
