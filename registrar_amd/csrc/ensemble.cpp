@@ -1,8 +1,10 @@
 // ensemble.cpp — synthetic in-process ZooKeeper ensemble (see ensemble.hpp).
 //
-// Threading model (the scaling core of the whole framework): one epoll loop
-// THREAD PER SERVER for parallel socket I/O, and SHARDED state so concurrent
-// registrar processes (which write disjoint domain subtrees) never contend:
+// Threading model (the scaling core of the whole framework): an IO-loop
+// thread POOL (connections assigned round-robin, independent of which
+// server socket accepted them — like ZooKeeper's selector threads) for
+// parallel socket I/O, and SHARDED state so concurrent registrar processes
+// (which write disjoint domain subtrees) never contend:
 //
 //   - znodes + their watch maps live in 64 hash shards, each under its own
 //     mutex; an op locks only the shard(s) of the paths it touches (create/

@@ -13,8 +13,10 @@
 //   - per-response latency injection,
 //   - one-shot data/child watches (Binder-style readers).
 //
-// Everything runs on one EventLoop thread; the control/introspection API is
-// thread-safe (tree guarded by a mutex, control posted to the loop).
+// Threading: an IO-loop pool (like ZooKeeper's selector threads) with
+// connections assigned round-robin, over 64-way sharded tree/watch state —
+// see ensemble.cpp's header comment for the full model and lock order. The
+// control/introspection API is thread-safe.
 #pragma once
 
 #include <atomic>
