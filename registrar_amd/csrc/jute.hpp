@@ -9,10 +9,13 @@
 //
 // Records implemented: ConnectRequest/Response, RequestHeader/ReplyHeader,
 // Stat, Create/Delete/Exists/GetData/SetData/GetChildren requests+responses,
-// WatcherEvent. Ops covered: create(±EPHEMERAL|SEQUENCE), delete, exists,
-// getData, setData, getChildren, ping, closeSession — the exact client verb
-// set the reference uses (stat/put/create/mkdirp/unlink/close)
-// plus watches so the synthetic ensemble can serve Binder-style readers.
+// MultiHeader, SetWatchesRequest, WatcherEvent. Ops covered:
+// create(±EPHEMERAL|SEQUENCE), delete, exists, getData, setData,
+// getChildren(+2), sync, multi, setWatches, ping, closeSession — the exact
+// client verb set the reference uses (stat/put/create/mkdirp/unlink/close,
+// SURVEY §2.4) plus watches and transactions so the synthetic ensemble can
+// serve Binder-style readers and the atomic-swap registration mode
+// (docs/protocol.md is the wire reference).
 #pragma once
 
 #include <cstdint>
