@@ -12,6 +12,8 @@ Subcommands:
   binder   --servers HOST:PORT[,...] [--port N]  serve DNS A/SRV answers from
                                                the registration tree
                                                (binder_lite)
+  verify   -f CONFIG                           pre-deployment smoke: connect,
+                                               register, heartbeat, unregister
 """
 import argparse
 import json
@@ -145,6 +147,56 @@ def cmd_binder(args):
     return 0
 
 
+def cmd_verify(args):
+    """Register → heartbeat → unregister once against the real ensemble in
+    the config, then exit. Safe: leaves no state behind (its ephemerals are
+    removed explicitly and die with the session regardless)."""
+    import registrar_amd as ra
+
+    with open(args.file) as f:
+        cfg = json.load(f)
+    try:
+        ra.Orchestrator(json.dumps(cfg))  # schema validation up front
+    except RuntimeError as e:
+        print(json.dumps({"verify": "invalid-config", "error": str(e)}), file=sys.stderr)
+        return 1
+    zk = cfg["zookeeper"]
+    client = ra.ZkClient(
+        servers=[(s["host"], int(s["port"])) for s in zk["servers"]],
+        session_timeout_ms=int(zk.get("timeout", 30000)),
+        connect_timeout_ms=int(zk.get("connectTimeout", 4000)),
+        connect_max_attempts=args.attempts,
+    )
+    client.start()
+    t0 = time.monotonic()
+    if not client.wait_connected(args.timeout * 1000):
+        print(json.dumps({"verify": "connect-failed"}), file=sys.stderr)
+        client.close()
+        return 1
+    connect_ms = round((time.monotonic() - t0) * 1000, 1)
+    reg = dict(cfg["registration"])
+    if cfg.get("adminIp") and not reg.get("adminIp"):
+        reg["adminIp"] = cfg["adminIp"]
+    rc, err, znodes = ra.register_node(client, json.dumps(reg))
+    if rc != 0:
+        print(json.dumps({"verify": "register-failed", "error": err}), file=sys.stderr)
+        client.close()
+        return 1
+    hb_rc, rtt_us = client.heartbeat(znodes)
+    un_rc = ra.unregister_node(client, znodes)
+    client.close()
+    ok = hb_rc == 0 and un_rc == 0
+    print(json.dumps({
+        "verify": "ok" if ok else "failed",
+        "connect_ms": connect_ms,
+        "znodes": znodes,
+        "heartbeat_rtt_ms": round(rtt_us / 1000.0, 3),
+        "heartbeat": ra.error_name(hb_rc),
+        "unregister": ra.error_name(un_rc),
+    }))
+    return 0 if ok else 1
+
+
 def cmd_check(args):
     import registrar_amd as ra
 
@@ -186,6 +238,12 @@ def main(argv=None):
     ck = sub.add_parser("check", help="validate a config file")
     ck.add_argument("-f", "--file", required=True)
     ck.set_defaults(fn=cmd_check)
+
+    v = sub.add_parser("verify", help="pre-deployment registration smoke test")
+    v.add_argument("-f", "--file", required=True)
+    v.add_argument("--timeout", type=int, default=30, help="connect wait seconds")
+    v.add_argument("--attempts", type=int, default=5, help="connect attempts")
+    v.set_defaults(fn=cmd_verify)
 
     b = sub.add_parser("binder", help="serve DNS A/SRV answers from the registration tree")
     b.add_argument("--servers", required=True, help="ZK host:port[,host:port...]")

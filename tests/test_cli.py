@@ -84,3 +84,31 @@ def test_ensemble_subcommand():
         proc.send_signal(signal.SIGTERM)
         proc.wait(timeout=15)
     assert proc.returncode == 0
+
+
+def test_verify_subcommand(ensemble, tmp_path):
+    cfg = orch_config(ensemble, {"domain": "v.test", "type": "host", "hostname": "vh",
+                                 "adminIp": "127.0.0.1", "settleMs": 0})
+    p = tmp_path / "cfg.json"
+    p.write_text(json.dumps(cfg))
+    r = run_cli("verify", "-f", str(p))
+    assert r.returncode == 0, r.stdout + r.stderr
+    out = json.loads(r.stdout)
+    assert out["verify"] == "ok"
+    assert out["znodes"] == ["/test/v/vh"]
+    assert out["heartbeat_rtt_ms"] > 0
+    # nothing left behind
+    assert not ensemble.get("/test/v/vh")["exists"]
+
+
+def test_verify_unreachable(tmp_path):
+    from conftest import free_port
+
+    cfg = {"registration": {"domain": "x.y", "type": "host"},
+           "zookeeper": {"servers": [{"host": "127.0.0.1", "port": free_port()}],
+                         "timeout": 2000, "connectTimeout": 200}}
+    p = tmp_path / "cfg.json"
+    p.write_text(json.dumps(cfg))
+    r = run_cli("verify", "-f", str(p), "--timeout", "10", "--attempts", "2")
+    assert r.returncode == 1
+    assert "connect-failed" in r.stderr
