@@ -209,6 +209,7 @@ HealthRecord HealthCheck::evaluate(const ExecResult& res) {
   } else if (cfg_.stdout_match) {
     auto syn = std::regex::ECMAScript;
     if (cfg_.stdout_match->flags.find('i') != std::string::npos) syn |= std::regex::icase;
+    if (cfg_.stdout_match->flags.find('m') != std::string::npos) syn |= std::regex::multiline;
     bool matched = false;
     try {
       std::regex re(cfg_.stdout_match->pattern, syn);

@@ -166,3 +166,11 @@ def test_periodic_runner():
 def test_config_validation():
     with pytest.raises(RuntimeError, match="command"):
         ra.HealthCheck(json.dumps({"interval": 10}))
+
+
+def test_stdout_match_flags_multiline():
+    # 'm' flag: ^/$ match per line (ECMAScript multiline)
+    h = hc({"command": "printf 'noise\\nOK\\nnoise\\n'", "stdoutMatch": {"pattern": "^OK$", "flags": "m"}})
+    assert h.check_once()["type"] == "ok"
+    h2 = hc({"command": "printf 'noise\\nOK\\nnoise\\n'", "stdoutMatch": {"pattern": "^OK$"}})
+    assert h2.check_once()["type"] == "fail"  # without m, ^$ anchor whole text
