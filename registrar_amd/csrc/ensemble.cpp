@@ -1122,10 +1122,17 @@ struct Ensemble::Impl {
           s->ephemerals.erase(path);
       }
       if (mine && !mine->alive.load(std::memory_order_acquire)) {
-        // our session died mid-txn: roll our created ephemerals back
-        std::lock_guard<std::mutex> eg(mine->eph_mu);
-        for (auto& [path, added, owner] : eph_changes)
-          if (added && owner == mine->id && mine->ephemerals.erase(path)) delete_node(path, mine);
+        // our session died mid-txn: roll our created ephemerals back.
+        // Collect under eph_mu, delete after releasing it — delete_node's
+        // owner bookkeeping re-locks the same eph_mu (self-deadlock
+        // otherwise; caught by the chaos stress watchdog).
+        std::vector<std::string> rollback;
+        {
+          std::lock_guard<std::mutex> eg(mine->eph_mu);
+          for (auto& [path, added, owner] : eph_changes)
+            if (added && owner == mine->id && mine->ephemerals.erase(path)) rollback.push_back(path);
+        }
+        for (const auto& path : rollback) delete_node(path, mine);
       }
     }
 
@@ -1445,7 +1452,7 @@ struct Ensemble::Impl {
     if (c->outbuf.empty()) {
       size_t off = 0;
       while (off < pkt.size()) {
-        ssize_t n = write(c->fd, pkt.data() + off, pkt.size() - off);
+        ssize_t n = send(c->fd, pkt.data() + off, pkt.size() - off, MSG_NOSIGNAL);
         if (n > 0) {
           off += static_cast<size_t>(n);
         } else if (n < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) {
@@ -1492,7 +1499,7 @@ struct Ensemble::Impl {
       std::lock_guard<std::mutex> og(c->out_mu);
       size_t off = 0;
       while (off < c->outbuf.size()) {
-        ssize_t n = write(c->fd, c->outbuf.data() + off, c->outbuf.size() - off);
+        ssize_t n = send(c->fd, c->outbuf.data() + off, c->outbuf.size() - off, MSG_NOSIGNAL);
         if (n > 0) {
           off += static_cast<size_t>(n);
         } else if (n < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) {

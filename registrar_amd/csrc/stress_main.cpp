@@ -9,6 +9,12 @@
 //     cycles with watches armed,
 //   - a chaos thread expiring random sessions and kill/restarting servers.
 // Exit code 0 = every cycle behaved; sanitizers report races/leaks.
+#include <dirent.h>
+#include <execinfo.h>
+#include <signal.h>
+#include <sys/syscall.h>
+#include <unistd.h>
+
 #include <atomic>
 #include <chrono>
 #include <cstdio>
@@ -22,9 +28,49 @@
 
 using namespace registrar;
 
+namespace {
+// Hang diagnosis: STRESS_WATCHDOG=<sec> arms a watchdog that, if the run
+// overshoots, SIGUSR1s every thread (whose handler prints a backtrace) and
+// aborts. Diagnostic aid; off by default.
+void dump_backtrace(int) {
+  void* frames[48];
+  int n = backtrace(frames, 48);
+  char hdr[64];
+  int len = snprintf(hdr, sizeof(hdr), "--- thread %ld backtrace ---\n", syscall(SYS_gettid));
+  ssize_t r = write(2, hdr, len);
+  (void)r;
+  backtrace_symbols_fd(frames, n, 2);
+}
+
+void arm_watchdog(int seconds) {
+  struct sigaction sa;
+  memset(&sa, 0, sizeof(sa));
+  sa.sa_handler = dump_backtrace;
+  sigaction(SIGUSR1, &sa, nullptr);
+  std::thread([seconds] {
+    std::this_thread::sleep_for(std::chrono::seconds(seconds));
+    fprintf(stderr, "WATCHDOG: run overshot %ds; dumping all threads\n", seconds);
+    DIR* d = opendir("/proc/self/task");
+    if (d) {
+      while (struct dirent* e = readdir(d)) {
+        if (e->d_name[0] == '.') continue;
+        long tid = atol(e->d_name);
+        syscall(SYS_tgkill, getpid(), tid, SIGUSR1);
+        std::this_thread::sleep_for(std::chrono::milliseconds(150));
+      }
+      closedir(d);
+    }
+    std::this_thread::sleep_for(std::chrono::seconds(1));
+    _exit(42);
+  }).detach();
+}
+}  // namespace
+
 int main(int argc, char** argv) {
+  signal(SIGPIPE, SIG_IGN);  // belt: library writes already use MSG_NOSIGNAL
   int nclients = 4;
   int seconds = 8;
+  if (const char* wd = getenv("STRESS_WATCHDOG")) arm_watchdog(atoi(wd));
   for (int i = 1; i < argc; i++) {
     if (!strcmp(argv[i], "-c") && i + 1 < argc) nclients = atoi(argv[++i]);
     if (!strcmp(argv[i], "-t") && i + 1 < argc) seconds = atoi(argv[++i]);

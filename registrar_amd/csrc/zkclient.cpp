@@ -529,7 +529,7 @@ struct ZkClient::Impl {
     if (fd < 0) return;
     size_t off = 0;
     while (off < outbuf.size()) {
-      ssize_t n = write(fd, outbuf.data() + off, outbuf.size() - off);
+      ssize_t n = send(fd, outbuf.data() + off, outbuf.size() - off, MSG_NOSIGNAL);
       if (n > 0) {
         off += static_cast<size_t>(n);
       } else if (n < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) {
