@@ -123,3 +123,44 @@ def test_gpu_payload_does_not_break_answers(ensemble, client, binder):
     assert an == 1
     _, _, rdata = parse_answers(buf, an)[0]
     assert struct.unpack(">HHH", rdata[:6])[2] == 8000
+
+
+def test_full_triangle_health_drives_dns(ensemble, binder, tmp_path):
+    """The complete discovery loop: an orchestrated registrar serves DNS via
+    binder-lite; a health failure takes it out of DNS; recovery brings it
+    back — the end-to-end behavior the reference system delivers with three
+    separate components (registrar + ZooKeeper + Binder)."""
+    import registrar_amd as ra
+    from conftest import orch_config, wait_for
+
+    gate = tmp_path / "gate"
+    gate.write_text("")
+    cfg = orch_config(
+        ensemble,
+        {"domain": "tri.dns.test", "type": "host", "adminIp": "10.9.0.1", "hostname": "t0",
+         "settleMs": 0,
+         "service": {"type": "service", "service": {"srvce": "_svc", "proto": "_tcp", "port": 9000}}},
+        heartbeatInterval=100,
+        healthCheck={"command": "test -e %s" % gate, "interval": 40, "timeout": 500,
+                     "threshold": 2, "period": 60000},
+    )
+    o = ra.Orchestrator(json.dumps(cfg))
+    o.start()
+    assert o.wait_registered(15000)
+
+    def answers(qtype=1):
+        buf, rcode, an = dns_query(binder.address, "tri.dns.test", qtype)
+        return an
+
+    assert wait_for(lambda: answers() == 1, timeout=5)
+    # SRV answer carries the service port
+    buf, rcode, an = dns_query(binder.address, "tri.dns.test", 33)
+    assert an == 1
+    assert struct.unpack(">HHH", parse_answers(buf, an)[0][2][:6])[2] == 9000
+
+    gate.unlink()  # GPU/instance goes sick → flap damping → unregister
+    assert wait_for(lambda: answers() == 0, timeout=10)
+
+    gate.write_text("")  # recovery → re-register → back in DNS
+    assert wait_for(lambda: answers() == 1, timeout=10)
+    o.stop()
