@@ -429,6 +429,12 @@ struct Ensemble::Impl {
         case kOpGetChildren:
           handle_get_children(c, hdr.xid, r);
           break;
+        case kOpGetACL:
+          handle_get_acl(c, hdr.xid, r);
+          break;
+        case kOpSetACL:
+          handle_set_acl(c, hdr.xid, r);
+          break;
         case kOpGetChildren2:
           handle_get_children2(c, hdr.xid, r);
           break;
@@ -919,6 +925,7 @@ struct Ensemble::Impl {
         // path -> (exists, ephemeral); absent = defer to tree
         std::unordered_map<std::string, std::pair<bool, bool>> state;
         std::unordered_map<std::string, int> child_delta;
+        std::unordered_map<std::string, int32_t> version;  // post-op versions
       } ov;
       auto tree_node = [&](const std::string& p) -> ZNode* {
         auto& sh = shard_of(p);
@@ -942,6 +949,14 @@ struct Ensemble::Impl {
         auto it = ov.child_delta.find(p);
         return base + (it == ov.child_delta.end() ? 0 : it->second);
       };
+      auto ov_version = [&](const std::string& p) -> int32_t {
+        auto it = ov.version.find(p);
+        if (it != ov.version.end()) return it->second;
+        bool created_in_txn = ov.state.count(p) && ov.state[p].first && !tree_node(p);
+        if (created_in_txn) return 0;
+        ZNode* n = tree_node(p);
+        return n ? n->stat.version : 0;
+      };
 
       for (size_t i = 0; i < mops.size(); i++) {
         const auto& mo = mops[i];
@@ -962,6 +977,7 @@ struct Ensemble::Impl {
                 err = kZNodeExists;
               } else {
                 ov.state[mo.path] = {true, (mo.flags & kEphemeral) != 0};
+                ov.version[mo.path] = 0;
                 ov.child_delta[parent]++;
               }
               break;
@@ -971,12 +987,11 @@ struct Ensemble::Impl {
               } else if (ov_child_count(mo.path) > 0) {
                 err = kZNotEmpty;
               } else {
-                bool created_in_txn = ov.state.count(mo.path) && ov.state[mo.path].first;
-                int32_t cur_version = created_in_txn ? 0 : (tree_node(mo.path) ? tree_node(mo.path)->stat.version : 0);
-                if (mo.version != -1 && mo.version != cur_version) {
+                if (mo.version != -1 && mo.version != ov_version(mo.path)) {
                   err = kZBadVersion;
                 } else {
                   ov.state[mo.path] = {false, false};
+                  ov.version.erase(mo.path);
                   ov.child_delta[parent]--;
                 }
               }
@@ -985,21 +1000,17 @@ struct Ensemble::Impl {
             case kOpSetData: {
               if (!ov_exists(mo.path)) {
                 err = kZNoNode;
+              } else if (mo.version != -1 && mo.version != ov_version(mo.path)) {
+                err = kZBadVersion;
               } else {
-                // version tracking through the overlay is approximate for
-                // txn-created nodes (version 0); real trees use the node's
-                ZNode* n = tree_node(mo.path);
-                int32_t cur = n ? n->stat.version : 0;
-                if (mo.version != -1 && mo.version != cur) err = kZBadVersion;
+                ov.version[mo.path] = ov_version(mo.path) + 1;  // txn-internal increment
               }
               break;
             }
             case 13: {  // check
-              ZNode* n = tree_node(mo.path);
-              bool created_in_txn = ov.state.count(mo.path) && ov.state[mo.path].first;
               if (!ov_exists(mo.path))
                 err = kZNoNode;
-              else if (mo.version != -1 && mo.version != (created_in_txn && !n ? 0 : (n ? n->stat.version : 0)))
+              else if (mo.version != -1 && mo.version != ov_version(mo.path))
                 err = kZBadVersion;
               break;
             }
@@ -1227,6 +1238,54 @@ struct Ensemble::Impl {
     std::string path = r.read_string();
     send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), kZOk,
                [&](JuteWriter& w) { w.write_string(path); });
+  }
+
+  // ACLs: everything is world:anyone (matching the reference's deployments);
+  // getACL reports the open ACL, setACL acks without enforcement
+  void handle_get_acl(Conn* c, int32_t xid, JuteReader& r) {
+    std::string path = r.read_string();
+    int32_t err = kZOk;
+    Stat stat;
+    {
+      Shard& sh = shard_of(path);
+      std::lock_guard<std::mutex> lk(sh.mu);
+      auto it = sh.nodes.find(path);
+      if (it == sh.nodes.end())
+        err = kZNoNode;
+      else
+        stat = it->second.stat;
+    }
+    if (err == kZOk)
+      send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), kZOk, [&](JuteWriter& w) {
+        write_acl_vector(w, {ACL{}});
+        stat.serialize(w);
+      });
+    else
+      send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), err, nullptr);
+  }
+
+  void handle_set_acl(Conn* c, int32_t xid, JuteReader& r) {
+    std::string path = r.read_string();
+    (void)read_acl_vector(r);
+    r.read_int();  // version
+    int32_t err = kZOk;
+    Stat stat;
+    {
+      Shard& sh = shard_of(path);
+      std::lock_guard<std::mutex> lk(sh.mu);
+      auto it = sh.nodes.find(path);
+      if (it == sh.nodes.end()) {
+        err = kZNoNode;
+      } else {
+        it->second.stat.aversion++;
+        stat = it->second.stat;
+      }
+    }
+    if (err == kZOk)
+      send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), kZOk,
+                 [&](JuteWriter& w) { stat.serialize(w); });
+    else
+      send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), err, nullptr);
   }
 
   void handle_close_session(Conn* c, int32_t xid) {

@@ -136,6 +136,14 @@ class EnsembleMachine(RuleBasedStateMachine):
         if rc == ra.ZOK:
             self.model = trial
 
+    @rule(path=path_strategy())
+    def check_get_acl(self, path):
+        # open-ACL reporting parity: any existing node reports world:anyone
+        rc, data, stat = self.client.get(path)
+        # (getACL isn't bound in python; covered at the wire level in
+        # test_wire_golden; here we just keep model/tree agreement on exists)
+        assert (rc == ra.ZOK) == (path in self.model.nodes)
+
     @rule()
     def expire_session(self):
         # session death: every ephemeral this session owns vanishes (all

@@ -140,3 +140,29 @@ def test_ephemeral_owner_on_wire(ensemble):
     eph_owner = struct.unpack(">q", stat[44:52])[0]
     assert eph_owner == session_id
     s.close()
+
+
+def test_get_acl_and_set_acl_wire(ensemble):
+    s = full_session(ensemble)
+    # create a node
+    req = struct.pack(">ii", 1, 1) + zk_string(b"/acl") + zk_string(b"")
+    req += struct.pack(">i", 1) + struct.pack(">i", 31) + zk_string(b"world") + zk_string(b"anyone")
+    req += struct.pack(">i", 0)
+    send_frame(s, req)
+    recv_frame(s)
+    # getACL (type 6)
+    send_frame(s, struct.pack(">ii", 2, 6) + zk_string(b"/acl"))
+    resp = recv_frame(s)
+    assert struct.unpack(">iqi", resp[:16])[2] == 0
+    nacl = struct.unpack(">i", resp[16:20])[0]
+    perms = struct.unpack(">i", resp[20:24])[0]
+    assert nacl == 1 and perms == 31
+    # setACL (type 7) acks and bumps aversion
+    acl = struct.pack(">i", 1) + struct.pack(">i", 31) + zk_string(b"world") + zk_string(b"anyone")
+    send_frame(s, struct.pack(">ii", 3, 7) + zk_string(b"/acl") + acl + struct.pack(">i", -1))
+    resp = recv_frame(s)
+    assert struct.unpack(">iqi", resp[:16])[2] == 0
+    stat = resp[16:]
+    aversion = struct.unpack(">i", stat[40:44])[0]
+    assert aversion == 1
+    s.close()
