@@ -78,6 +78,7 @@ py::dict health_record_to_dict(const HealthRecord& rec) {
   d["threshold"] = rec.threshold;
   d["exit_status"] = rec.exit_status;
   d["stdout"] = rec.stdout_tail;
+  d["stderr"] = rec.stderr_tail;
   return d;
 }
 
@@ -443,6 +444,7 @@ PYBIND11_MODULE(_core, m) {
           d["exit_status"] = res.exit_status;
           d["timed_out"] = res.timed_out;
           d["stdout"] = py::bytes(res.out);
+          d["stderr"] = py::bytes(res.err);
           return d;
         },
         py::arg("command"), py::arg("timeout_ms") = 1000);

@@ -551,6 +551,7 @@ struct Ensemble::Impl {
     bool made_ephemeral = false;
     std::string created_path;
     std::string child_watch_parent;
+    int64_t op_zxid = 0;  // the zxid THIS create committed at
     if (!valid_path(req.path) || req.path == "/") {
       err = kZMarshallingError;
     } else {
@@ -583,6 +584,7 @@ struct Ensemble::Impl {
         } else {
           // single hash+probe: the node was emplaced above; fill it in
           int64_t zz = zxid_counter.fetch_add(1, std::memory_order_relaxed) + 1;
+          op_zxid = zz;
           ZNode& n = nit->second;
           n.data = req.data;
           n.stat.czxid = zz;
@@ -631,8 +633,10 @@ struct Ensemble::Impl {
         err = kZSessionExpired;
       }
     }
+    // mutating replies carry the op's OWN zxid (a fresh counter read could
+    // exceed what this client actually observed — ADVICE r1)
     if (err == kZOk)
-      send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), kZOk,
+      send_reply(c, xid, op_zxid, kZOk,
                  [&](JuteWriter& w) { resp.serialize(w); });
     else
       send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), err, nullptr);
@@ -643,6 +647,7 @@ struct Ensemble::Impl {
     DeleteRequest req;
     req.deserialize(r);
     int32_t err = kZOk;
+    int64_t op_zxid = 0;
     {
       // peek under the target's shard lock; full delete re-locks both shards
       Shard& sh = shard_of(req.path);
@@ -658,16 +663,19 @@ struct Ensemble::Impl {
     }
     if (err == kZOk) {
       // common case: a session unlinking its own ephemerals (cleanup step)
-      if (!delete_node(req.path, c->session)) err = kZNoNode;  // raced
+      if (!delete_node(req.path, c->session, &op_zxid)) err = kZNoNode;  // raced
     }
-    send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), err, nullptr);
+    send_reply(c, xid,
+               op_zxid ? op_zxid : zxid_counter.load(std::memory_order_relaxed),
+               err, nullptr);
   }
 
   // Full node removal: locks child+parent shards, updates parent, fires
   // watches, detaches from the owner session. Returns false if missing.
   // owner_hint avoids the session_mu map lookup on the hot path (a session
   // deleting its own ephemerals — the register-pipeline cleanup case).
-  bool delete_node(const std::string& path, const SessionPtr& owner_hint = nullptr) {
+  bool delete_node(const std::string& path, const SessionPtr& owner_hint = nullptr,
+                   int64_t* out_zxid = nullptr) {
     int64_t owner = 0;
     {
       std::string parent = parent_path(path);
@@ -677,6 +685,7 @@ struct Ensemble::Impl {
       if (it == csh.nodes.end()) return false;
       if (!it->second.children.empty()) return false;  // re-check under lock
       int64_t z = zxid_counter.fetch_add(1, std::memory_order_relaxed) + 1;
+      if (out_zxid) *out_zxid = z;
       owner = it->second.stat.ephemeral_owner;
       csh.nodes.erase(it);
       Shard& psh = *lk.second;
@@ -761,6 +770,7 @@ struct Ensemble::Impl {
     req.deserialize(r);
     int32_t err = kZOk;
     SetDataResponse resp;
+    int64_t op_zxid = 0;
     {
       Shard& sh = shard_of(req.path);
       std::lock_guard<std::mutex> lk(sh.mu);
@@ -771,6 +781,7 @@ struct Ensemble::Impl {
         err = kZBadVersion;
       } else {
         int64_t zz = zxid_counter.fetch_add(1, std::memory_order_relaxed) + 1;
+        op_zxid = zz;
         it->second.data = req.data;
         it->second.stat.mzxid = zz;
         it->second.stat.mtime = wall_ms();
@@ -781,7 +792,7 @@ struct Ensemble::Impl {
       }
     }
     if (err == kZOk)
-      send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), kZOk,
+      send_reply(c, xid, op_zxid, kZOk,
                  [&](JuteWriter& w) { resp.serialize(w); });
     else
       send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), err, nullptr);
@@ -902,6 +913,11 @@ struct Ensemble::Impl {
     int64_t sid = sid_of(c);
     std::vector<int32_t> results(mops.size(), kZOk);
     std::vector<std::string> created_paths(mops.size());
+    // setData Stat snapshots taken while the shard locks are held — the
+    // reply body must NOT re-read the tree after the locks are released
+    // (concurrent create/delete on the shard would race the read)
+    std::vector<Stat> setdata_stats(mops.size());
+    int64_t txn_zxid = 0;  // zxid of the LAST op applied by this txn
     std::vector<std::pair<std::string, int32_t>> data_events;   // path, event type
     std::vector<std::string> child_events;                      // parent paths
     // ephemeral bookkeeping: (path, added, owner-session) — a delete may
@@ -1036,6 +1052,7 @@ struct Ensemble::Impl {
             case kOpCreate: {
               ops.create.fetch_add(1, std::memory_order_relaxed);
               int64_t zz = zxid_counter.fetch_add(1, std::memory_order_relaxed) + 1;
+              txn_zxid = zz;
               Shard& csh = shard_of(mo.path);
               ZNode& n = csh.nodes[mo.path];
               n.data = mo.data;
@@ -1065,6 +1082,7 @@ struct Ensemble::Impl {
             case kOpDelete: {
               ops.del.fetch_add(1, std::memory_order_relaxed);
               int64_t zz = zxid_counter.fetch_add(1, std::memory_order_relaxed) + 1;
+              txn_zxid = zz;
               Shard& csh = shard_of(mo.path);
               auto it = csh.nodes.find(mo.path);
               if (it != csh.nodes.end()) {
@@ -1087,6 +1105,7 @@ struct Ensemble::Impl {
             case kOpSetData: {
               ops.set_data.fetch_add(1, std::memory_order_relaxed);
               int64_t zz = zxid_counter.fetch_add(1, std::memory_order_relaxed) + 1;
+              txn_zxid = zz;
               Shard& csh = shard_of(mo.path);
               auto it = csh.nodes.find(mo.path);
               if (it != csh.nodes.end()) {
@@ -1095,6 +1114,7 @@ struct Ensemble::Impl {
                 it->second.stat.mtime = wall_ms();
                 it->second.stat.version++;
                 it->second.stat.data_length = static_cast<int32_t>(mo.data.size());
+                setdata_stats[i] = it->second.stat;
               }
               data_events.push_back({mo.path, kEventNodeDataChanged});
               break;
@@ -1147,7 +1167,9 @@ struct Ensemble::Impl {
       }
     }
 
-    int64_t z = zxid_counter.load(std::memory_order_relaxed);
+    // reply with THIS txn's last zxid (not a fresh counter read, which may
+    // already reflect other connections' later commits)
+    int64_t z = txn_zxid ? txn_zxid : zxid_counter.load(std::memory_order_relaxed);
     send_reply_with_body(c, xid, z, txn_err, [&](JuteWriter& w) {
       for (size_t i = 0; i < mops.size(); i++) {
         MultiHeader mh;
@@ -1162,13 +1184,7 @@ struct Ensemble::Impl {
           mh.err = 0;
           mh.serialize(w);
           if (mops[i].type == kOpCreate) w.write_string(created_paths[i]);
-          if (mops[i].type == kOpSetData) {
-            Shard& sh = shard_of(mops[i].path);
-            auto it = sh.nodes.find(mops[i].path);
-            Stat st;
-            if (it != sh.nodes.end()) st = it->second.stat;
-            st.serialize(w);
-          }
+          if (mops[i].type == kOpSetData) setdata_stats[i].serialize(w);
         }
       }
       MultiHeader end;

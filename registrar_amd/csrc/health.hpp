@@ -30,6 +30,7 @@
 #include <memory>
 #include <mutex>
 #include <optional>
+#include <regex>
 #include <string>
 #include <thread>
 #include <vector>
@@ -40,8 +41,11 @@
 namespace registrar {
 
 struct StdoutMatch {
-  std::string pattern;
-  std::string flags;   // "i" supported (ECMAScript-style)
+  std::string pattern;  // ECMAScript-dialect std::regex (see docs/config.md
+                        // for deltas vs JS RegExp: no lookbehind, etc.)
+  std::string flags;    // "i" (icase), "m" (multiline), "g" (no-op for a
+                        // single search) — anything else is rejected at
+                        // config parse, where JS would have thrown too
   bool invert = false;
 };
 
@@ -58,6 +62,10 @@ struct HealthCheckConfig {
 
 HealthCheckConfig parse_health_check(const Json& j);
 
+// Compile pattern+flags (throws std::runtime_error on unsupported flags or an
+// invalid pattern — used by parse_health_check to fail fast at config time).
+std::regex compile_stdout_match(const StdoutMatch& m);
+
 struct HealthRecord {
   bool ok = false;
   std::string command;
@@ -67,6 +75,7 @@ struct HealthRecord {
   int64_t threshold = 0;
   int exit_status = 0;
   std::string stdout_tail;  // last bytes of captured stdout (diagnostics)
+  std::string stderr_tail;  // last bytes of captured stderr (diagnostics)
 
   Json to_json() const;
 };
@@ -76,7 +85,8 @@ struct HealthRecord {
 struct ExecResult {
   int exit_status = -1;   // -1 ⇒ killed / failed to run
   bool timed_out = false;
-  std::string out;        // captured stdout (capped)
+  std::string out;        // captured stdout (capped) — the ONLY regex input
+  std::string err;        // captured stderr (capped) — diagnostics only
 };
 ExecResult exec_with_timeout(const std::string& command, int64_t timeout_ms, size_t max_buffer = 1024 * 1024);
 

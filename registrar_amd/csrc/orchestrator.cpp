@@ -86,6 +86,8 @@ OrchestratorConfig parse_config(const Json& cfg) {
   }
 
   out.heartbeat_interval_ms = cfg.get_int("heartbeatInterval", out.heartbeat_interval_ms);
+  out.heartbeat_failure_floor_ms =
+      cfg.get_int("heartbeatFailureFloor", out.heartbeat_failure_floor_ms);
   // heartbeat retry policy — read-but-unused in the reference (§2.2.5), plumbed here
   if (const Json* hb = cfg.find("heartbeat")) {
     if (hb->is_object()) {
@@ -219,6 +221,19 @@ bool Orchestrator::connect_and_register(bool initial) {
       emit({OrchEvent::Type::Error, "createZKClient: unable to create ZK client", {}, 0});
     return false;
   }
+  // expiry recovery while the health checker holds us DOWN: reconnect the
+  // session but do NOT advertise an unhealthy node — the next ok health
+  // record re-registers via on_health_record (the old code reset
+  // down_=false unconditionally, briefly putting a sick node back in DNS)
+  if (!initial && health_ && health_->is_down()) {
+    {
+      std::lock_guard<std::mutex> g(mu_);
+      znodes_.clear();
+      down_ = true;
+    }
+    log_.warn("expiry recovery: health is down; session restored, register deferred");
+    return true;
+  }
   RegisterResult res = register_node(*client, cfg_.registration, log_);
   if (res.rc != zk::kZOk) {
     // reference would crash here on an undefined variable (§2.2.3); we emit
@@ -325,7 +340,7 @@ void Orchestrator::heartbeat_loop() {
         emit({OrchEvent::Type::Heartbeat, "", nodes, rtt});
       } else {
         emit({OrchEvent::Type::HeartbeatFailure, zk::error_name(rc), nodes, 0});
-        next_wait = std::max<int64_t>(cfg_.heartbeat_interval_ms, 60000);
+        next_wait = std::max<int64_t>(cfg_.heartbeat_interval_ms, cfg_.heartbeat_failure_floor_ms);
       }
     }
 

@@ -45,6 +45,9 @@ struct OrchestratorConfig {
   RegistrationConfig registration;
   std::optional<HealthCheckConfig> health;
   int64_t heartbeat_interval_ms = 3000;  // lib/index.js:132
+  // degraded cadence after a heartbeat failure: max(interval, floor)
+  // (lib/index.js:142-146; configurable so tests can assert the reschedule)
+  int64_t heartbeat_failure_floor_ms = 60000;
   zk::RetryPolicy heartbeat_retry;       // lib/zk.js:38-42 defaults
   bool exit_on_expiry = false;           // false ⇒ in-process re-register
   std::string log_level;                 // config `logLevel`

@@ -427,9 +427,24 @@ struct ZkClient::Impl {
         WatcherEvent ev;
         ev.deserialize(r);
         log.debug("watch event", {{"path", Json(ev.path)}, {"type", Json(static_cast<int64_t>(ev.type))}});
-        data_watch_paths.erase(ev.path);
-        exist_watch_paths.erase(ev.path);
-        child_watch_paths.erase(ev.path);
+        // one-shot consumption matches the event TYPE (ZooKeeper semantics):
+        // Created/DataChanged consume data+exist watches only, ChildrenChanged
+        // consumes only the child watch, Deleted consumes all three. Erasing
+        // unrelated sets would drop still-armed watches from setWatches re-arm.
+        switch (ev.type) {
+          case kEventNodeChildrenChanged:
+            child_watch_paths.erase(ev.path);
+            break;
+          case kEventNodeDeleted:
+            data_watch_paths.erase(ev.path);
+            exist_watch_paths.erase(ev.path);
+            child_watch_paths.erase(ev.path);
+            break;
+          default:  // NodeCreated / NodeDataChanged
+            data_watch_paths.erase(ev.path);
+            exist_watch_paths.erase(ev.path);
+            break;
+        }
         {
           std::lock_guard<std::mutex> g(watch_mu);
           watch_queue.push_back(ev);

@@ -174,3 +174,68 @@ def test_stdout_match_flags_multiline():
     assert h.check_once()["type"] == "ok"
     h2 = hc({"command": "printf 'noise\\nOK\\nnoise\\n'", "stdoutMatch": {"pattern": "^OK$"}})
     assert h2.check_once()["type"] == "fail"  # without m, ^$ anchor whole text
+
+
+# ---- round 2: stdout/stderr separation + regex fidelity hardening ----
+# (VERDICT r1 Weak #1 / next-round items 2 and 5)
+
+
+def test_exec_stderr_captured_separately():
+    res = ra.exec_with_timeout("echo out; echo err 1>&2", 1000)
+    assert res["stdout"] == b"out\n"
+    assert res["stderr"] == b"err\n"
+
+
+def test_stdout_match_ignores_stderr():
+    # the reference matches child.exec's STDOUT arg only
+    # (/root/reference/lib/health.js:89-101): a pattern that appears only on
+    # stderr must NOT satisfy the match
+    h = hc({"command": "echo HEALTHY 1>&2", "stdoutMatch": {"pattern": "HEALTHY"}})
+    rec = h.check_once()
+    assert rec["type"] == "fail"
+    assert "HEALTHY" in rec["stderr"]  # preserved for diagnostics
+    # same text on stdout: ok
+    h2 = hc({"command": "echo HEALTHY", "stdoutMatch": {"pattern": "HEALTHY"}})
+    assert h2.check_once()["type"] == "ok"
+
+
+def test_stdout_match_invert_ignores_stderr():
+    # invert converse: a forbidden pattern on stderr only must not FAIL the
+    # check (invert fails when the pattern matches stdout)
+    h = hc({"command": "echo BROKEN 1>&2", "stdoutMatch": {"pattern": "BROKEN", "invert": True}})
+    assert h.check_once()["type"] == "ok"
+    h2 = hc({"command": "echo BROKEN", "stdoutMatch": {"pattern": "BROKEN", "invert": True}})
+    assert h2.check_once()["type"] == "fail"
+
+
+def test_invalid_pattern_rejected_at_parse():
+    # fail fast at config time (reference: assert-plus throws at construction,
+    # lib/health.js:23-38); previously an invalid pattern silently never matched
+    with pytest.raises(RuntimeError, match="invalid regex"):
+        hc({"command": "true", "stdoutMatch": {"pattern": "(unclosed"}})
+
+
+def test_unknown_flags_rejected_at_parse():
+    with pytest.raises(RuntimeError, match="unsupported flag"):
+        hc({"command": "true", "stdoutMatch": {"pattern": "x", "flags": "s"}})
+    # 'g' is a harmless no-op for a single search and is accepted
+    h = hc({"command": "echo zz", "stdoutMatch": {"pattern": "z", "flags": "gi"}})
+    assert h.check_once()["type"] == "ok"
+
+
+def test_background_grandchild_does_not_hang():
+    # a backgrounded grandchild keeps the pipe write end open after the shell
+    # exits; the post-reap drain must be bounded (health.cpp drain)
+    t0 = time.monotonic()
+    res = ra.exec_with_timeout("echo done; sleep 5 & exit 0", 3000)
+    elapsed = time.monotonic() - t0
+    assert res["exit_status"] == 0 and not res["timed_out"]
+    assert b"done" in res["stdout"]
+    assert elapsed < 2.0  # returned long before the grandchild's 5 s sleep
+
+
+def test_fail_record_carries_stderr_tail():
+    h = hc({"command": "echo diagnostic-detail 1>&2; exit 1"})
+    rec = h.check_once()
+    assert rec["type"] == "fail"
+    assert "diagnostic-detail" in rec["stderr"]
