@@ -414,6 +414,9 @@ struct Ensemble::Impl {
         case kOpCreate:
           handle_create(c, hdr.xid, r);
           break;
+        case kOpCreate2:  // create + Stat in the response (3.5+ clients)
+          handle_create(c, hdr.xid, r, /*with_stat=*/true);
+          break;
         case kOpDelete:
           handle_delete(c, hdr.xid, r);
           break;
@@ -541,7 +544,7 @@ struct Ensemble::Impl {
 
   // --- ops ---
 
-  void handle_create(Conn* c, int32_t xid, JuteReader& r) {
+  void handle_create(Conn* c, int32_t xid, JuteReader& r, bool with_stat = false) {
     ops.create.fetch_add(1, std::memory_order_relaxed);
     CreateRequest req;
     req.deserialize(r);
@@ -551,7 +554,8 @@ struct Ensemble::Impl {
     bool made_ephemeral = false;
     std::string created_path;
     std::string child_watch_parent;
-    int64_t op_zxid = 0;  // the zxid THIS create committed at
+    int64_t op_zxid = 0;   // the zxid THIS create committed at
+    Stat created_stat;     // snapshot for create2 replies
     if (!valid_path(req.path) || req.path == "/") {
       err = kZMarshallingError;
     } else {
@@ -605,6 +609,7 @@ struct Ensemble::Impl {
           fire_child_watches_locked(psh, parent);
           resp.path = path;
           created_path = path;
+          created_stat = n.stat;
         }
       }
       (void)child_watch_parent;
@@ -636,8 +641,10 @@ struct Ensemble::Impl {
     // mutating replies carry the op's OWN zxid (a fresh counter read could
     // exceed what this client actually observed — ADVICE r1)
     if (err == kZOk)
-      send_reply(c, xid, op_zxid, kZOk,
-                 [&](JuteWriter& w) { resp.serialize(w); });
+      send_reply(c, xid, op_zxid, kZOk, [&](JuteWriter& w) {
+        resp.serialize(w);
+        if (with_stat) created_stat.serialize(w);  // Create2Response
+      });
     else
       send_reply(c, xid, zxid_counter.load(std::memory_order_relaxed), err, nullptr);
   }
