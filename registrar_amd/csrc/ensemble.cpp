@@ -704,7 +704,7 @@ struct Ensemble::Impl {
         pit->second.stat.num_children = static_cast<int32_t>(pit->second.children.size());
         fire_child_watches_locked(psh, parent);
       }
-      fire_data_watches_locked(csh, path, kEventNodeDeleted);
+      fire_deleted_watches_locked(csh, path);
     }
     if (owner != 0) {
       SessionPtr s;
@@ -927,6 +927,7 @@ struct Ensemble::Impl {
     int64_t txn_zxid = 0;  // zxid of the LAST op applied by this txn
     std::vector<std::pair<std::string, int32_t>> data_events;   // path, event type
     std::vector<std::string> child_events;                      // parent paths
+    std::vector<std::string> deleted_paths;                     // NodeDeleted → own child watchers
     // ephemeral bookkeeping: (path, added, owner-session) — a delete may
     // remove ANOTHER session's ephemeral; its owner's set must be updated
     std::vector<std::tuple<std::string, bool, int64_t>> eph_changes;
@@ -1107,6 +1108,7 @@ struct Ensemble::Impl {
               }
               data_events.push_back({mo.path, kEventNodeDeleted});
               child_events.push_back(parent);
+              deleted_paths.push_back(mo.path);
               break;
             }
             case kOpSetData: {
@@ -1132,7 +1134,9 @@ struct Ensemble::Impl {
         }
         // fire watches while holding the shard locks (same discipline as
         // the single-op handlers: shard → session_mu → conns_mu → out_mu)
-        for (auto& [path, ev] : data_events) fire_data_watches_locked(shard_of(path), path, ev);
+        for (auto& [path, ev] : data_events)
+          if (ev != kEventNodeDeleted) fire_data_watches_locked(shard_of(path), path, ev);
+        for (auto& path : deleted_paths) fire_deleted_watches_locked(shard_of(path), path);
         std::set<std::string> fired;
         for (auto& parent : child_events)
           if (fired.insert(parent).second) fire_child_watches_locked(shard_of(parent), parent);
@@ -1389,6 +1393,23 @@ struct Ensemble::Impl {
     std::set<int64_t> watchers = std::move(it->second);
     sh.child_watches.erase(it);
     deliver_watch(watchers, path, kEventNodeChildrenChanged);
+  }
+
+  // NodeDeleted goes to data watchers AND child watchers of the deleted
+  // node, deduplicated per session — ZooKeeper's DataTree passes the
+  // already-triggered set to the child-watch trigger so a session holding
+  // both watch kinds gets ONE event
+  void fire_deleted_watches_locked(Shard& sh, const std::string& path) {
+    std::set<int64_t> watchers;
+    if (auto it = sh.data_watches.find(path); it != sh.data_watches.end()) {
+      watchers = std::move(it->second);
+      sh.data_watches.erase(it);
+    }
+    if (auto it = sh.child_watches.find(path); it != sh.child_watches.end()) {
+      watchers.insert(it->second.begin(), it->second.end());
+      sh.child_watches.erase(it);
+    }
+    if (!watchers.empty()) deliver_watch(watchers, path, kEventNodeDeleted);
   }
 
   void deliver_watch(const std::set<int64_t>& watchers, const std::string& path, int32_t event_type) {

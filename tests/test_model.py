@@ -3,8 +3,13 @@ reference model of ZooKeeper's tree semantics (hypothesis stateful test).
 
 Covers create (±ephemeral), delete, setData, exists, getData, getChildren,
 with version checks, parent/child rules, and error codes — every operation's
-result is cross-checked against the model."""
+result is cross-checked against the model. Watch semantics are modeled too
+(VERDICT r1 #9): exists-watch on a nonexistent node fired by create, one-shot
+consumption, NodeDeleted union-delivery to data+child watchers, child events
+on the parent, and watch death across session expiry."""
 import json
+import time
+from collections import Counter
 
 import pytest
 from hypothesis import HealthCheck, settings
@@ -83,6 +88,12 @@ class EnsembleMachine(RuleBasedStateMachine):
         self.ens = ra.Ensemble(servers=1, tick_ms=200)
         self.ens.start()
         self.model = Model()
+        # watch model: server-side data watches (exists/getData arm the same
+        # set), child watches, expected undelivered events, received events
+        self.watched_data = set()
+        self.watched_child = set()
+        self.expected_events = Counter()
+        self.received = Counter()
         self._new_client()
 
     def _new_client(self):
@@ -96,11 +107,54 @@ class EnsembleMachine(RuleBasedStateMachine):
             self.client.close()
             self.ens.stop()
 
+    # ---- watch-event bookkeeping (ZooKeeper delivery rules) ----
+
+    def _note_create(self, path):
+        if path in self.watched_data:  # exist-watch on a nonexistent node
+            self.watched_data.discard(path)
+            self.expected_events[("created", path)] += 1
+        par = Model.parent(path)
+        if par in self.watched_child:
+            self.watched_child.discard(par)
+            self.expected_events[("child", par)] += 1
+
+    def _note_delete(self, path):
+        # NodeDeleted: ONE event even if both data and child watches are armed
+        if path in self.watched_data or path in self.watched_child:
+            self.watched_data.discard(path)
+            self.watched_child.discard(path)
+            self.expected_events[("deleted", path)] += 1
+        par = Model.parent(path)
+        if par in self.watched_child:
+            self.watched_child.discard(par)
+            self.expected_events[("child", par)] += 1
+
+    def _note_set(self, path):
+        if path in self.watched_data:
+            self.watched_data.discard(path)
+            self.expected_events[("changed", path)] += 1
+
+    def _settle_watches(self):
+        """Wait until every expected event arrived, then require exact
+        multiset equality — a spurious or missing event fails here."""
+        deadline = time.monotonic() + 5
+        while time.monotonic() < deadline:
+            for w in self.client.poll_watches():
+                kind = w["type"]
+                self.received[(kind, w["path"])] += 1
+            if self.received == self.expected_events:
+                return
+            time.sleep(0.01)
+        assert self.received == self.expected_events, \
+            "watch mismatch: got %r want %r" % (dict(self.received), dict(self.expected_events))
+
     @rule(path=path_strategy(), data=st.sampled_from(DATAS), ephemeral=st.booleans())
     def create(self, path, data, ephemeral):
         rc, _ = self.client.create(path, data, ephemeral)
         expect = self.model.create(path, data, ephemeral)
         assert rc == expect, "create %s: got %s want %s" % (path, ra.error_name(rc), ra.error_name(expect))
+        if rc == ra.ZOK:
+            self._note_create(path)
 
     @rule(path=path_strategy(), version=st.sampled_from([-1, 0, 1, 7]))
     def delete(self, path, version):
@@ -108,12 +162,41 @@ class EnsembleMachine(RuleBasedStateMachine):
         expect = self.model.delete(path, version)
         assert rc == expect, "delete %s v%d: got %s want %s" % (path, version, ra.error_name(rc),
                                                                ra.error_name(expect))
+        if rc == ra.ZOK:
+            self._note_delete(path)
 
     @rule(path=path_strategy(), data=st.sampled_from(DATAS), version=st.sampled_from([-1, 0, 1, 7]))
     def set_data(self, path, data, version):
         rc = self.client.set(path, data, version)
         expect = self.model.set(path, data, version)
         assert rc == expect
+        if rc == ra.ZOK:
+            self._note_set(path)
+
+    # ---- watch arming rules ----
+
+    @rule(path=path_strategy())
+    def arm_exist_watch(self, path):
+        # an exist watch arms even when the node does NOT exist (it fires on
+        # a later create) — lib consumers (Binder) rely on this
+        self.client.exists(path, watch=True)
+        self.watched_data.add(path)
+
+    @rule(path=path_strategy())
+    def arm_data_watch(self, path):
+        rc, _, _ = self.client.get(path, watch=True)
+        if rc == ra.ZOK:  # getData arms only on an existing node
+            self.watched_data.add(path)
+
+    @rule(path=path_strategy())
+    def arm_child_watch(self, path):
+        rc, _ = self.client.get_children(path, watch=True)
+        if rc == ra.ZOK:
+            self.watched_child.add(path)
+
+    @rule()
+    def check_watch_delivery(self):
+        self._settle_watches()
 
     @rule(ops=st.lists(
         st.tuples(st.sampled_from(["create", "delete"]), st.lists(st.sampled_from(NAMES), min_size=1,
@@ -135,6 +218,12 @@ class EnsembleMachine(RuleBasedStateMachine):
         assert rc == expect, "multi: got %s want %s (%r)" % (ra.error_name(rc), ra.error_name(expect), mops)
         if rc == ra.ZOK:
             self.model = trial
+            # watches fire per applied op, in order (rollback fires nothing)
+            for kind, path, data, eph in mops:
+                if kind == "create":
+                    self._note_create(path)
+                else:
+                    self._note_delete(path)
 
     @rule(path=path_strategy())
     def check_get_acl(self, path):
@@ -151,9 +240,16 @@ class EnsembleMachine(RuleBasedStateMachine):
         # terminal and must be replaced, exactly like a daemon would
         import time
 
+        # drain in-flight watch events first: whatever fires AFTER this point
+        # belongs to the dying session and must never reach the new one
+        self._settle_watches()
         self.ens.expire_session(self.client.session_id())
         for p in [p for p, v in self.model.nodes.items() if v[2]]:
             del self.model.nodes[p]
+        # watches do NOT survive into a new session (ZooKeeper semantics):
+        # armed watches die with the session, with no event owed
+        self.watched_data.clear()
+        self.watched_child.clear()
         deadline = time.monotonic() + 10
         while time.monotonic() < deadline and self.client.state() != "expired":
             time.sleep(0.02)
