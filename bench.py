@@ -44,6 +44,67 @@ def percentile(values, p):
     return vs[idx]
 
 
+def read_cpu_quota():
+    """Effective CPU quota (fractional CPUs) from cgroup v2 or v1; None = unlimited."""
+    try:  # v2
+        q, p = open("/sys/fs/cgroup/cpu.max").read().split()
+        if q != "max":
+            return float(q) / float(p)
+    except (OSError, ValueError):
+        pass
+    try:  # v1
+        q = int(open("/sys/fs/cgroup/cpu/cpu.cfs_quota_us").read())
+        p = int(open("/sys/fs/cgroup/cpu/cpu.cfs_period_us").read())
+        if q > 0:
+            return q / p
+    except (OSError, ValueError):
+        pass
+    return None
+
+
+def read_cpu_stat():
+    """CFS throttle counters (v2 usec / v1 ns, normalized to usec); None if absent."""
+    for path in ("/sys/fs/cgroup/cpu.stat", "/sys/fs/cgroup/cpu/cpu.stat"):
+        try:
+            d = {}
+            for line in open(path):
+                parts = line.split()
+                if len(parts) == 2:
+                    d[parts[0]] = int(parts[1])
+        except OSError:
+            continue
+        if "throttled_usec" in d:  # v2
+            return {"nr_throttled": d.get("nr_throttled", 0),
+                    "throttled_usec": d["throttled_usec"],
+                    "usage_usec": d.get("usage_usec")}
+        if "throttled_time" in d:  # v1 (ns)
+            return {"nr_throttled": d.get("nr_throttled", 0),
+                    "throttled_usec": d["throttled_time"] // 1000,
+                    "usage_usec": None}
+    return None
+
+
+def partition_cpus(world, rank):
+    """Partition the allowed CPUs into per-rank client slices plus an
+    ensemble slice (rank 0 hosts the ensemble's IO pool). Returns
+    (my_cpus, ensemble_cpus, effective_cpu_count) — my_cpus/ensemble_cpus are
+    None when partitioning is pointless (single proc or too few CPUs)."""
+    cpus = sorted(os.sched_getaffinity(0))
+    quota = read_cpu_quota()
+    effective = len(cpus) if quota is None else max(1, min(len(cpus), int(quota)))
+    if world <= 1 or os.environ.get("BENCH_AFFINITY", "1") == "0":
+        return None, None, effective
+    usable = cpus[:effective]
+    # world+1 slots: one per rank + one (the remainder too) for the ensemble;
+    # with fewer usable CPUs than slots, pinning only adds contention — skip
+    if len(usable) < world + 1:
+        return None, None, effective
+    k = len(usable) // (world + 1)
+    my = set(usable[rank * k:(rank + 1) * k])
+    ens = set(usable[world * k:])  # ensemble gets the last slot + remainder
+    return my, ens, effective
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -88,6 +149,15 @@ def main():
         if use_cuda:
             torch.cuda.synchronize()
 
+    # ---- CPU topology: per-rank affinity + quota-sized ensemble IO pool ----
+    # Under a CFS quota (the r1 lease throttled the 4-proc point into a
+    # non-monotone curve) free-roaming threads over-subscribe the cgroup and
+    # burn the budget on context switches. Pin each rank's client threads to
+    # its own CPU slice and give the ensemble IO pool a dedicated slice; the
+    # timed region is bracketed by cpu.stat reads so any residual throttling
+    # self-labels in the output JSON. BENCH_AFFINITY=0 disables pinning.
+    my_cpus, ens_cpus, effective_cpus = partition_cpus(world, rank)
+
     # ---- ensemble: hosted by rank 0, shared over localhost TCP ----
     if args.servers <= 0:
         # production ZK quorum sizes: 1 standalone / 3 / 5 (BASELINE configs
@@ -96,13 +166,22 @@ def main():
     ensemble = None
     if rank == 0:
         io_threads = int(os.environ.get("BENCH_IO_THREADS", "0"))
+        if io_threads == 0 and ens_cpus:
+            io_threads = max(2, len(ens_cpus))  # size the pool to its slice
+        if ens_cpus:
+            os.sched_setaffinity(0, ens_cpus)  # IO threads inherit this mask
         ensemble = ra.Ensemble(servers=args.servers, tick_ms=100, max_session_timeout_ms=60000,
                                io_threads=io_threads)
         ensemble.start()
         connect = ensemble.connect_string()
-        log("rank0 hosts ensemble at %s" % connect)
+        log("rank0 hosts ensemble at %s (io_threads=%d, ens_cpus=%s)"
+            % (connect, io_threads, sorted(ens_cpus) if ens_cpus else "all"))
     else:
         connect = None
+    if my_cpus:
+        # client/worker threads created from here on inherit the rank slice
+        os.sched_setaffinity(0, my_cpus)
+        log("rank %d pinned to cpus %s" % (rank, sorted(my_cpus)))
     if dist is not None:
         obj = [connect]
         dist.broadcast_object_list(obj, src=0)
@@ -170,9 +249,11 @@ def main():
 
     # ---- timed region ----
     barrier()
+    stat_before = read_cpu_stat() if rank == 0 else None
     t0 = time.perf_counter()
     rtts_us = [step() for _ in range(args.steps)]
     elapsed = time.perf_counter() - t0
+    stat_after = read_cpu_stat() if rank == 0 else None
     barrier()
 
     # max elapsed over ranks is THE job time; p50 over all ranks' heartbeats
@@ -215,6 +296,18 @@ def main():
             "data": "synthetic",
             "p50_heartbeat_rtt_ms": round(p50_ms, 3),
             "p99_heartbeat_rtt_ms": round(p99_ms, 3),
+            # CFS-throttle evidence for the timed region (cgroup is shared by
+            # all ranks, so rank 0's delta covers the whole job): a throttled
+            # scaling point self-labels instead of silently bending the curve
+            "cpu_stat": {
+                "cpu_quota": read_cpu_quota(),
+                "cpus_allowed": effective_cpus,
+                "affinity": "per-rank slices + dedicated ensemble slice" if my_cpus else "none",
+                "nr_throttled_delta": (stat_after["nr_throttled"] - stat_before["nr_throttled"])
+                if stat_before and stat_after else None,
+                "throttled_usec_delta": (stat_after["throttled_usec"] - stat_before["throttled_usec"])
+                if stat_before and stat_after else None,
+            },
             "config": {
                 "model": "registrar re-register cycle, 1k ephemeral znodes/proc",
                 "global_batch": world * args.znodes,

@@ -117,7 +117,8 @@ class Logger {
 
   void log(LogLevel lvl, const std::string& msg) const { log(lvl, msg, {}); }
 
-  void log(LogLevel lvl, const std::string& msg, std::vector<JsonMember> fields) const {
+  void log(LogLevel lvl, const std::string& msg, std::vector<JsonMember> fields,
+           const char* file = nullptr, int line = 0) const {
     if (!enabled(lvl)) return;
     Json rec = Json::object();
     rec.set("name", Json(core_->name));
@@ -128,20 +129,53 @@ class Logger {
     rec.set("level", Json(static_cast<int64_t>(lvl)));
     rec.set("msg", Json(msg));
     rec.set("time", Json(iso8601_now()));
+    // bunyan `src:true` parity: when the logger runs at debug/trace
+    // verbosity, every record carries the caller's file:line
+    // (reference: main.js:75-76)
+    if (file && core_->level.load() <= static_cast<int>(LogLevel::Debug)) {
+      const char* base = file;
+      for (const char* p = file; *p; p++)
+        if (*p == '/') base = p + 1;
+      Json src = Json::object();
+      src.set("file", Json(std::string(base)));
+      src.set("line", Json(static_cast<int64_t>(line)));
+      rec.set("src", std::move(src));
+    }
     rec.set("v", Json(static_cast<int64_t>(0)));
-    std::string line = rec.dump();
-    line += '\n';
+    std::string line_out = rec.dump();
+    line_out += '\n';
     std::lock_guard<std::mutex> g(core_->mu);
-    fwrite(line.data(), 1, line.size(), core_->stream);
+    fwrite(line_out.data(), 1, line_out.size(), core_->stream);
     fflush(core_->stream);
   }
 
-  void trace(const std::string& msg, std::vector<JsonMember> f = {}) const { log(LogLevel::Trace, msg, std::move(f)); }
-  void debug(const std::string& msg, std::vector<JsonMember> f = {}) const { log(LogLevel::Debug, msg, std::move(f)); }
-  void info(const std::string& msg, std::vector<JsonMember> f = {}) const { log(LogLevel::Info, msg, std::move(f)); }
-  void warn(const std::string& msg, std::vector<JsonMember> f = {}) const { log(LogLevel::Warn, msg, std::move(f)); }
-  void error(const std::string& msg, std::vector<JsonMember> f = {}) const { log(LogLevel::Error, msg, std::move(f)); }
-  void fatal(const std::string& msg, std::vector<JsonMember> f = {}) const { log(LogLevel::Fatal, msg, std::move(f)); }
+  // __builtin_FILE/__builtin_LINE default args evaluate at the CALL site
+  // (gcc and ROCm clang both support them in C++17), giving bunyan-style
+  // src without a macro layer
+  void trace(const std::string& msg, std::vector<JsonMember> f = {},
+             const char* file = __builtin_FILE(), int line = __builtin_LINE()) const {
+    log(LogLevel::Trace, msg, std::move(f), file, line);
+  }
+  void debug(const std::string& msg, std::vector<JsonMember> f = {},
+             const char* file = __builtin_FILE(), int line = __builtin_LINE()) const {
+    log(LogLevel::Debug, msg, std::move(f), file, line);
+  }
+  void info(const std::string& msg, std::vector<JsonMember> f = {},
+            const char* file = __builtin_FILE(), int line = __builtin_LINE()) const {
+    log(LogLevel::Info, msg, std::move(f), file, line);
+  }
+  void warn(const std::string& msg, std::vector<JsonMember> f = {},
+            const char* file = __builtin_FILE(), int line = __builtin_LINE()) const {
+    log(LogLevel::Warn, msg, std::move(f), file, line);
+  }
+  void error(const std::string& msg, std::vector<JsonMember> f = {},
+             const char* file = __builtin_FILE(), int line = __builtin_LINE()) const {
+    log(LogLevel::Error, msg, std::move(f), file, line);
+  }
+  void fatal(const std::string& msg, std::vector<JsonMember> f = {},
+             const char* file = __builtin_FILE(), int line = __builtin_LINE()) const {
+    log(LogLevel::Fatal, msg, std::move(f), file, line);
+  }
 
  private:
   std::shared_ptr<LogCore> core_;

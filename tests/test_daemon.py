@@ -148,3 +148,37 @@ def test_daemon_sigusr1_metrics(ensemble, daemon_bin):
     assert metrics[0]["registers"] >= 1
     assert metrics[0]["heartbeats"] >= 1
     assert metrics[0]["p50HeartbeatRttUs"] > 0
+
+
+def test_debug_logs_carry_src(ensemble, daemon_bin):
+    """bunyan src parity (reference main.js:75-76): at debug verbosity every
+    record carries the emitting file:line (VERDICT r1 next-round #8)."""
+    cfg = orch_config(
+        ensemble,
+        {"domain": "src.test", "type": "host", "hostname": "sh", "settleMs": 0},
+        heartbeatInterval=100,
+        logLevel="debug",
+    )
+    proc, cfg_path = spawn_daemon(daemon_bin, cfg)
+    try:
+        time.sleep(1.0)
+    finally:
+        proc.send_signal(signal.SIGTERM)
+    recs = read_logs(proc)
+    os.unlink(cfg_path)
+    assert recs, "no log records"
+    with_src = [r for r in recs if "src" in r]
+    assert with_src, "no record carries src at debug level"
+    s = with_src[0]["src"]
+    assert s["file"].endswith(".cpp") or s["file"].endswith(".hpp")
+    assert isinstance(s["line"], int) and s["line"] > 0
+    # at default (info) verbosity src is absent
+    cfg.pop("logLevel")
+    proc, cfg_path = spawn_daemon(daemon_bin, cfg)
+    try:
+        time.sleep(0.8)
+    finally:
+        proc.send_signal(signal.SIGTERM)
+    recs = read_logs(proc)
+    os.unlink(cfg_path)
+    assert recs and all("src" not in r for r in recs)

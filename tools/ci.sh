@@ -1,7 +1,9 @@
 #!/bin/bash
-# CI pipeline (the reference's Jenkinsfile check→test equivalent):
-# build the extension + binaries, run the CPU test suite, then optional
-# sanitizer passes (tools/sanitize.sh).
+# CI pipeline (the reference's Jenkinsfile check→test equivalent,
+# /root/reference/Jenkinsfile:24-50): build the extension + binaries, run the
+# CPU test suite, run the concurrency stress under TSan/ASan (reduced chaos
+# iteration count), then a bench smoke. SKIP_SANITIZERS=1 skips the sanitizer
+# stage (e.g. on hosts without ROCm's clang).
 set -euo pipefail
 cd "$(dirname "$0")/.."
 
@@ -11,6 +13,13 @@ make daemon
 
 echo "== tests (CPU) =="
 python3 -m pytest tests/ -x -q -m "not gpu" --timeout=240 --timeout-method=thread
+
+if [ "${SKIP_SANITIZERS:-0}" != "1" ] && [ -x /opt/rocm/lib/llvm/bin/clang++ ]; then
+    echo "== sanitizers (TSan + ASan/UBSan chaos stress) =="
+    make tsan asan
+else
+    echo "== sanitizers: SKIPPED (no ROCm clang or SKIP_SANITIZERS=1) =="
+fi
 
 echo "== bench smoke =="
 timeout 300 python3 bench.py --gpus 1 --steps 5 --warmup 1 >/dev/null
