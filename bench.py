@@ -88,20 +88,30 @@ def partition_cpus(world, rank):
     """Partition the allowed CPUs into per-rank client slices plus an
     ensemble slice (rank 0 hosts the ensemble's IO pool). Returns
     (my_cpus, ensemble_cpus, effective_cpu_count) — my_cpus/ensemble_cpus are
-    None when partitioning is pointless (single proc or too few CPUs)."""
+    None when partitioning is pointless (single proc or too few CPUs).
+    BENCH_RANK_CPUS=k forces k CPUs per rank (remainder to the ensemble);
+    BENCH_AFFINITY=0 disables pinning entirely."""
     cpus = sorted(os.sched_getaffinity(0))
     quota = read_cpu_quota()
     effective = len(cpus) if quota is None else max(1, min(len(cpus), int(quota)))
-    if world <= 1 or os.environ.get("BENCH_AFFINITY", "1") == "0":
+    if os.environ.get("BENCH_AFFINITY", "1") == "0":
         return None, None, effective
     usable = cpus[:effective]
-    # world+1 slots: one per rank + one (the remainder too) for the ensemble;
-    # with fewer usable CPUs than slots, pinning only adds contention — skip
-    if len(usable) < world + 1:
+    k = int(os.environ.get("BENCH_RANK_CPUS", "0"))
+    if k > 0:
+        if len(usable) < world * k + 1:
+            return None, None, effective
+        my = set(usable[rank * k:(rank + 1) * k])
+        ens = set(usable[world * k:])
+        return my, ens, effective
+    # default: a registrar client is essentially one busy thread + the python
+    # driver — 2 CPUs per rank is plenty; everything left feeds the ensemble
+    # IO pool, which serves ALL ranks' traffic and is the scaling bottleneck
+    if world <= 1 or len(usable) < world + 2:
         return None, None, effective
-    k = len(usable) // (world + 1)
+    k = max(1, min(2, (len(usable) - 2) // world))
     my = set(usable[rank * k:(rank + 1) * k])
-    ens = set(usable[world * k:])  # ensemble gets the last slot + remainder
+    ens = set(usable[world * k:])
     return my, ens, effective
 
 

@@ -12,6 +12,10 @@ sys.path.insert(0, REPO_ROOT)
 
 def pytest_configure(config):
     config.addinivalue_line("markers", "gpu: test requires an AMD GPU (run on an MI355X box)")
+    config.addinivalue_line(
+        "markers",
+        "zk_real: interop test against a REAL Apache ZooKeeper at $ZK_HOST:$ZK_PORT "
+        "(skipped when unset; no JVM exists in the build container)")
 
 
 @pytest.fixture
