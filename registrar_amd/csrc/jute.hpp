@@ -210,6 +210,19 @@ class JuteReader {
 
   std::string read_buffer() { return read_string(); }
 
+  // Count prefix of a jute vector, bounded by what the frame can actually
+  // hold (each element is ≥ min_elem bytes): a hostile count like 2^29 must
+  // throw immediately instead of allocating/constructing for minutes and
+  // wedging the IO loop (found by tests/test_fuzz.py hostile-frame cases).
+  int32_t read_vec_count(size_t min_elem) {
+    int32_t n = read_int();
+    if (n < 0) return 0;  // null vector
+    if (min_elem == 0) min_elem = 1;
+    if (static_cast<size_t>(n) > remaining() / min_elem)
+      throw std::runtime_error("jute: vector count exceeds frame");
+    return n;
+  }
+
  private:
   void need(size_t n) const {
     if (pos_ + n > len_) throw std::runtime_error("jute: short read");
@@ -374,7 +387,8 @@ inline void write_acl_vector(JuteWriter& w, const std::vector<ACL>& acls) {
 }
 
 inline std::vector<ACL> read_acl_vector(JuteReader& r) {
-  int32_t n = r.read_int();
+  // each ACL is ≥ 12 bytes (perms + two length-prefixed strings)
+  int32_t n = r.read_vec_count(12);
   std::vector<ACL> acls;
   if (n > 0) {
     acls.resize(static_cast<size_t>(n));
@@ -516,8 +530,9 @@ struct GetChildrenResponse {
     for (const auto& c : children) w.write_string(c);
   }
   void deserialize(JuteReader& r) {
-    int32_t n = r.read_int();
+    int32_t n = r.read_vec_count(4);  // each child name: 4-byte length prefix
     children.clear();
+    children.reserve(static_cast<size_t>(n));
     for (int32_t i = 0; i < n; i++) children.push_back(r.read_string());
   }
 };
@@ -558,8 +573,9 @@ struct SetWatchesRequest {
     for (const auto& s : v) w.write_string(s);
   }
   static std::vector<std::string> read_vec(JuteReader& r) {
-    int32_t n = r.read_int();
+    int32_t n = r.read_vec_count(4);  // each path: 4-byte length prefix
     std::vector<std::string> v;
+    v.reserve(static_cast<size_t>(n));
     for (int32_t i = 0; i < n; i++) v.push_back(r.read_string());
     return v;
   }

@@ -4,6 +4,7 @@ Usage: python setup.py build_ext --inplace
 The standalone daemon/ensemble binaries are built by the Makefile from the
 same csrc/ sources.
 """
+import glob
 import os
 
 from pybind11.setup_helpers import Pybind11Extension, build_ext
@@ -22,6 +23,9 @@ ext = Pybind11Extension(
         os.path.join(CSRC, "orchestrator.cpp"),
         os.path.join(CSRC, "gpu.cpp"),
     ],
+    # a header edit must rebuild every TU (setuptools skips unchanged .cpp
+    # otherwise — a jute.hpp fix once shipped stale objects)
+    depends=sorted(glob.glob(os.path.join(CSRC, "*.hpp"))),
     cxx_std=17,
     extra_compile_args=["-O2", "-g", "-Wall", "-pthread"],
     extra_link_args=["-pthread"],

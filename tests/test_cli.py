@@ -112,3 +112,21 @@ def test_verify_unreachable(tmp_path):
     r = run_cli("verify", "-f", str(p), "--timeout", "10", "--attempts", "2")
     assert r.returncode == 1
     assert "connect-failed" in r.stderr
+
+
+def test_check_rejects_bad_stdout_match(tmp_path):
+    """Fail-fast regex validation at config time (VERDICT r1 next-round #5)."""
+    base = {"zookeeper": {"servers": [{"host": "127.0.0.1", "port": 2181}]},
+            "registration": {"domain": "a.b", "type": "host"}}
+    bad_pat = dict(base, healthCheck={"command": "true", "stdoutMatch": {"pattern": "(unclosed"}})
+    p = tmp_path / "badpat.json"
+    p.write_text(json.dumps(bad_pat))
+    r = run_cli("check", "-f", str(p))
+    assert r.returncode == 1 and "invalid regex" in r.stderr
+
+    bad_flag = dict(base, healthCheck={"command": "true",
+                                       "stdoutMatch": {"pattern": "x", "flags": "su"}})
+    p2 = tmp_path / "badflag.json"
+    p2.write_text(json.dumps(bad_flag))
+    r = run_cli("check", "-f", str(p2))
+    assert r.returncode == 1 and "unsupported flag" in r.stderr

@@ -91,3 +91,50 @@ def test_fuzz_random_streams(ensemble):
     rc, _ = c.create("/post-fuzz", b"ok")
     assert rc == ra.ZOK
     c.close()
+
+
+def test_malformed_multi_and_negative_lengths(ensemble):
+    """Hostile jute bodies post-handshake: negative string lengths, absurd
+    ACL counts, truncated multi framing — each closes only the offending
+    connection; the ensemble keeps serving."""
+    cases = [
+        # create with negative path length
+        struct.pack(">ii", 1, 1) + struct.pack(">i", -5),
+        # create with path length far beyond the frame
+        struct.pack(">ii", 1, 1) + struct.pack(">i", 1 << 30) + b"/x",
+        # create with absurd ACL vector count
+        struct.pack(">ii", 1, 1) + struct.pack(">i", 2) + b"/y"
+        + struct.pack(">i", 0) + struct.pack(">i", 1 << 29),
+        # multi with op body truncated mid-header
+        struct.pack(">ii", 1, 14) + struct.pack(">i", 1),
+        # multi that never sends the done terminator (frame just ends)
+        struct.pack(">ii", 1, 14) + struct.pack(">i?i", 13, False, -1)
+        + struct.pack(">i", 2) + b"/z" + struct.pack(">i", -1),
+        # setWatches with a negative vector count
+        struct.pack(">ii", -8, 101) + struct.pack(">q", 0) + struct.pack(">i", -3),
+    ]
+    for body in cases:
+        s = raw_conn(ensemble)
+        handshake(s)
+        s.sendall(struct.pack(">i", len(body)) + body)
+        s.settimeout(2)
+        try:
+            s.recv(64)  # server may reply with an error or just close
+        except socket.timeout:
+            pass
+        s.close()
+        assert ensemble_alive(ensemble)
+
+
+def test_zero_and_negative_frame_lengths(ensemble):
+    for framelen in (0, -1, -(1 << 31)):
+        s = raw_conn(ensemble)
+        handshake(s)
+        s.sendall(struct.pack(">i", framelen))
+        s.settimeout(2)
+        try:
+            s.recv(16)
+        except socket.timeout:
+            pass
+        s.close()
+        assert ensemble_alive(ensemble)
