@@ -509,3 +509,39 @@ def test_set_watches_golden(ensemble):
     assert struct.unpack(">ii", ev[16:24]) == (3, 3)
     a2.close()
     b.close()
+
+
+def test_self_watch_event_precedes_triggering_reply(ensemble):
+    """ZooKeeper ordering guarantee: when a client's own operation fires its
+    own watch, the WatcherEvent frame is delivered BEFORE the reply to the
+    operation that triggered it (both ride the same ordered outbox)."""
+    s = full_session(ensemble)
+    send_frame(s, create_req(1, b"/selfw", b"v0"))
+    assert reply_header(recv_frame(s))[2] == 0
+    send_frame(s, struct.pack(">ii", 2, 4) + zk_string(b"/selfw") + b"\x01")  # getData+watch
+    assert reply_header(recv_frame(s))[2] == 0
+    send_frame(s, struct.pack(">ii", 3, 5) + zk_string(b"/selfw") + zk_string(b"v1")
+               + struct.pack(">i", -1))
+    first = recv_frame(s)
+    second = recv_frame(s)
+    assert reply_header(first)[0] == -1, "watch event must precede the setData reply"
+    assert struct.unpack(">ii", first[16:24]) == (3, 3)  # NodeDataChanged
+    assert reply_header(second)[:1] == (3,) and reply_header(second)[2] == 0
+    s.close()
+
+
+def test_session_timeout_negotiation_golden(ensemble):
+    # requested timeout is clamped into [min, max] and echoed in the
+    # ConnectResponse (real ZK: 2×tick .. 20×tick; ours: configurable floor)
+    s = connect_raw(ensemble)
+    send_frame(s, struct.pack(">iqiq", 0, 0, 1, 0) + zk_string(b"\x00" * 16))  # absurdly small
+    resp = recv_frame(s)
+    lo = struct.unpack(">i", resp[4:8])[0]
+    assert lo > 1  # negotiated UP to the server floor
+    s.close()
+    s = connect_raw(ensemble)
+    send_frame(s, struct.pack(">iqiq", 0, 0, 10**9, 0) + zk_string(b"\x00" * 16))  # absurdly large
+    resp = recv_frame(s)
+    hi = struct.unpack(">i", resp[4:8])[0]
+    assert hi <= 60000  # capped at the server max
+    s.close()
