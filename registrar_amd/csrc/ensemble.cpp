@@ -453,6 +453,10 @@ struct Ensemble::Impl {
         case kOpCloseSession:
           handle_close_session(c, hdr.xid);
           break;
+        case 100:  // auth packet (xid -4): ACLs are open here, ack and ignore
+                   // so real third-party clients that always send auth work
+          send_reply(c, hdr.xid, zxid_counter.load(std::memory_order_relaxed), kZOk, nullptr);
+          break;
         default:
           ops.unknown.fetch_add(1, std::memory_order_relaxed);
           send_reply(c, hdr.xid, zxid_counter.load(std::memory_order_relaxed), kZSystemError, nullptr);

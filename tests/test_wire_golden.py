@@ -545,3 +545,34 @@ def test_session_timeout_negotiation_golden(ensemble):
     hi = struct.unpack(">i", resp[4:8])[0]
     assert hi <= 60000  # capped at the server max
     s.close()
+
+
+def test_sequence_create_golden(ensemble):
+    """SEQUENCE creates append a zero-padded 10-digit counter (parent
+    cversion), like real ZooKeeper's %010d suffix."""
+    s = full_session(ensemble)
+    send_frame(s, create_req(1, b"/seqparent"))
+    assert reply_header(recv_frame(s))[2] == 0
+    send_frame(s, create_req(2, b"/seqparent/n-", flags=2))  # SEQUENCE
+    resp = recv_frame(s)
+    assert reply_header(resp)[2] == 0
+    # counter = parent cversion at create time; a fresh parent starts at 0
+    assert resp[16:] == zk_string(b"/seqparent/n-0000000000")
+    send_frame(s, create_req(3, b"/seqparent/n-", flags=2))
+    resp = recv_frame(s)
+    assert resp[16:] == zk_string(b"/seqparent/n-0000000001")
+    s.close()
+
+
+def test_auth_packet_acked(ensemble):
+    """Real clients send an auth packet (xid -4, op 100) right after the
+    handshake; the open-ACL ensemble must ack it, not error the session."""
+    s = full_session(ensemble)
+    body = struct.pack(">ii", -4, 100) + struct.pack(">i", 0) + zk_string(b"digest") + zk_string(b"u:p")
+    send_frame(s, body)
+    resp = recv_frame(s)
+    assert reply_header(resp)[:1] == (-4,) and reply_header(resp)[2] == 0
+    # session still fully functional
+    send_frame(s, create_req(1, b"/after-auth"))
+    assert reply_header(recv_frame(s))[2] == 0
+    s.close()
