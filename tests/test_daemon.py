@@ -182,3 +182,23 @@ def test_debug_logs_carry_src(ensemble, daemon_bin):
     recs = read_logs(proc)
     os.unlink(cfg_path)
     assert recs and all("src" not in r for r in recs)
+
+
+def test_verbose_flag_steps_log_level(ensemble, daemon_bin):
+    """-v lowers the level one step per repeat (reference main.js:66-76:
+    bunyan level minus 10 each): info → debug with one -v, visible as debug
+    records (which also carry src at that point)."""
+    cfg = orch_config(
+        ensemble,
+        {"domain": "vflag.test", "type": "host", "hostname": "vh", "settleMs": 0},
+        heartbeatInterval=100,
+    )
+    proc, cfg_path = spawn_daemon(daemon_bin, cfg, "-v")
+    try:
+        time.sleep(0.9)
+    finally:
+        proc.send_signal(signal.SIGTERM)
+    recs = read_logs(proc)
+    os.unlink(cfg_path)
+    assert any(r["level"] == 20 for r in recs), "one -v must enable debug (level 20) records"
+    assert not any(r["level"] == 10 for r in recs), "one -v must not reach trace"

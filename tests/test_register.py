@@ -207,3 +207,16 @@ def test_prepared_registration_matches_register_node(ensemble, client):
     # idempotent re-register through the same prepared handle
     rc, err, znodes2 = prep.register_(client)
     assert rc == ra.ZOK and znodes2 == znodes
+
+
+def test_payload_address_fallback_without_adminip(ensemble, client):
+    """Without adminIp the payload advertises the first non-loopback IPv4
+    interface address (reference lib/register.js:22-31's address()); with no
+    such interface it falls back to 127.0.0.1 rather than crashing."""
+    import ipaddress
+
+    znodes = reg(client, {"domain": "noadmin.test", "type": "host", "hostname": "nf"})
+    obj = json.loads(ensemble.get(znodes[0])["data"])
+    addr = obj["address"]
+    ipaddress.IPv4Address(addr)  # must be a syntactically valid IPv4 literal
+    assert obj["host"]["address"] == addr
