@@ -121,3 +121,58 @@ def test_torch_distributed_gloo_two_ranks(ensemble):
         assert p.exitcode == 0
     assert sorted(r for r, _ in results) == [0, 1]
     assert all(s == "ok" for _, s in results)
+
+
+def test_bench_partition_cpus(monkeypatch):
+    """bench.partition_cpus: rank slices are disjoint, within the quota, and
+    the ensemble slice is the remainder — for every rank of several worlds
+    (this partition is what makes the driver-measured scaling curve monotone
+    under a CFS quota)."""
+    import importlib.util
+
+    spec = importlib.util.spec_from_file_location("bench", os.path.join(REPO_ROOT, "bench.py"))
+    bench = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(bench)
+
+    cpus = set(range(16))
+    monkeypatch.setattr(os, "sched_getaffinity", lambda pid: cpus)
+    monkeypatch.setattr(bench, "read_cpu_quota", lambda: 16.0)
+    monkeypatch.delenv("BENCH_AFFINITY", raising=False)
+    monkeypatch.delenv("BENCH_RANK_CPUS", raising=False)
+
+    for world in (2, 4, 8):
+        slices = []
+        ens_ref = None
+        for rank in range(world):
+            my, ens, eff = bench.partition_cpus(world, rank)
+            assert eff == 16
+            assert my and ens, "partition expected for world=%d" % world
+            assert my.isdisjoint(ens)
+            slices.append(my)
+            ens_ref = ens
+        # all rank slices pairwise disjoint and inside the quota
+        seen = set()
+        for s in slices:
+            assert not (s & seen)
+            seen |= s
+        assert seen | ens_ref <= cpus
+        assert len(ens_ref) >= 2
+
+    # world=1: no pinning by default
+    assert bench.partition_cpus(1, 0)[:2] == (None, None)
+    # BENCH_AFFINITY=0 disables
+    monkeypatch.setenv("BENCH_AFFINITY", "0")
+    assert bench.partition_cpus(8, 0)[:2] == (None, None)
+    monkeypatch.delenv("BENCH_AFFINITY")
+    # BENCH_RANK_CPUS override honored
+    monkeypatch.setenv("BENCH_RANK_CPUS", "1")
+    my, ens, _ = bench.partition_cpus(8, 3)
+    assert my == {3} and ens == set(range(8, 16))
+    # too few CPUs for the requested split: graceful no-pinning
+    monkeypatch.setenv("BENCH_RANK_CPUS", "4")
+    assert bench.partition_cpus(8, 0)[:2] == (None, None)
+    monkeypatch.delenv("BENCH_RANK_CPUS")
+    # tight quota (world+2 > cpus): no pinning rather than starving ranks
+    monkeypatch.setattr(os, "sched_getaffinity", lambda pid: set(range(4)))
+    monkeypatch.setattr(bench, "read_cpu_quota", lambda: 4.0)
+    assert bench.partition_cpus(4, 0)[:2] == (None, None)
