@@ -210,6 +210,15 @@ class JuteReader {
 
   std::string read_buffer() { return read_string(); }
 
+  // Validate-and-skip a length-prefixed string without materializing it
+  // (hot-path requests carry ACL scheme/id strings the server discards).
+  void skip_string() {
+    int32_t n = read_int();
+    if (n < 0) return;  // null string/buffer
+    need(static_cast<size_t>(n));
+    pos_ += static_cast<size_t>(n);
+  }
+
   // Count prefix of a jute vector, bounded by what the frame can actually
   // hold (each element is ≥ min_elem bytes): a hostile count like 2^29 must
   // throw immediately instead of allocating/constructing for minutes and
@@ -410,9 +419,17 @@ struct CreateRequest {
     w.write_int(flags);
   }
   void deserialize(JuteReader& r) {
+    // validate-and-skip the ACL vector instead of materializing it: the
+    // open-ACL server discards it, and a 1k-create burst would otherwise
+    // allocate 2 strings per op just to free them (hot path)
     path = r.read_string();
     data = r.read_buffer();
-    acls = read_acl_vector(r);
+    int32_t n = r.read_vec_count(12);
+    for (int32_t i = 0; i < n; i++) {
+      r.read_int();    // perms
+      r.skip_string();  // scheme
+      r.skip_string();  // id
+    }
     flags = r.read_int();
   }
 };
