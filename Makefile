@@ -9,9 +9,14 @@ CORE_SRCS := $(SRCDIR)/ensemble.cpp $(SRCDIR)/zkclient.cpp $(SRCDIR)/registrar.c
              $(SRCDIR)/health.cpp $(SRCDIR)/orchestrator.cpp $(SRCDIR)/gpu.cpp
 HDRS := $(wildcard $(SRCDIR)/*.hpp)
 
-.PHONY: all ext daemon test clean tsan asan stress
+.PHONY: all ext daemon test check clean tsan asan stress
 
 all: ext daemon
+
+# the reference's `make check` (lint+style CI gate, Jenkinsfile:24-50)
+# maps to the full local pipeline: build, CPU suite, sanitizers, bench smoke
+check:
+	bash tools/ci.sh
 
 ext:
 	python3 setup.py build_ext --inplace
