@@ -68,10 +68,10 @@ def main():
         gi = i % max(1, len(gpus)) if gpus else 0
         cfg = {
             "zookeeper": {"servers": servers, "timeout": 3000, "connectTimeout": 4000},
+            "gpuIndex": gi,  # top-level, like etc/config.gpu0.json
             "registration": {
                 "domain": DOMAIN, "type": "host", "adminIp": "10.88.0.%d" % (i + 1),
                 "hostname": "gpu%d" % i, "settleMs": 0, "ports": [9000 + i],
-                "gpuIndex": gi,
             },
             "heartbeatInterval": 200,
             "healthCheck": {"command": "gpu-liveness", "interval": 500,
