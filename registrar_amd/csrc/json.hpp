@@ -315,6 +315,9 @@ class Json {
     Json parse_value() {
       skip_ws();
       if (eof()) fail("unexpected end of input");
+      // recursive-descent depth cap: a hostile deeply-nested document must
+      // produce a parse error, not a stack overflow (SIGSEGV)
+      if (depth_ >= 256) fail("nesting too deep (max 256)");
       char c = s_[i_];
       switch (c) {
         case '{':
@@ -354,10 +357,12 @@ class Json {
 
     Json parse_object() {
       i_++;  // '{'
+      depth_++;
       Json obj = Json::object();
       skip_ws();
       if (!eof() && s_[i_] == '}') {
         i_++;
+        depth_--;
         return obj;
       }
       while (true) {
@@ -376,6 +381,7 @@ class Json {
         }
         if (s_[i_] == '}') {
           i_++;
+          depth_--;
           return obj;
         }
         fail("expected ',' or '}'");
@@ -384,10 +390,12 @@ class Json {
 
     Json parse_array() {
       i_++;  // '['
+      depth_++;
       Json arr = Json::array();
       skip_ws();
       if (!eof() && s_[i_] == ']') {
         i_++;
+        depth_--;
         return arr;
       }
       while (true) {
@@ -400,6 +408,7 @@ class Json {
         }
         if (s_[i_] == ']') {
           i_++;
+          depth_--;
           return arr;
         }
         fail("expected ',' or ']'");
@@ -507,6 +516,7 @@ class Json {
 
     const std::string& s_;
     size_t i_ = 0;
+    int depth_ = 0;  // current object/array nesting (capped in parse_value)
   };
 
   Type type_;

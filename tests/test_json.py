@@ -68,3 +68,14 @@ def test_payload_has_no_null_ttl():
     """ttl must be ABSENT (not null) when unset — JSON.stringify semantics."""
     rec = json.loads(ra.build_host_record(json.dumps({"domain": "a.b", "type": "host", "adminIp": "1.2.3.4"})))
     assert "ttl" not in rec
+
+
+def test_depth_cap_no_stack_overflow():
+    """A hostile deeply-nested document must raise a parse error, not
+    overflow the recursive-descent stack (previously SIGSEGV at ~100k deep)."""
+    deep = "[" * 100000 + "]" * 100000
+    with pytest.raises(RuntimeError, match="nesting too deep"):
+        ra.json_roundtrip(deep)
+    # legitimate nesting well below the cap still parses
+    ok = "[" * 200 + "0" + "]" * 200
+    assert ra.json_roundtrip(ok) == ok
