@@ -75,7 +75,7 @@ def test_depth_cap_no_stack_overflow():
     overflow the recursive-descent stack (previously SIGSEGV at ~100k deep)."""
     deep = "[" * 100000 + "]" * 100000
     with pytest.raises(RuntimeError, match="nesting too deep"):
-        ra.json_roundtrip(deep)
+        roundtrip(deep)
     # legitimate nesting well below the cap still parses
     ok = "[" * 200 + "0" + "]" * 200
-    assert ra.json_roundtrip(ok) == ok
+    assert roundtrip(ok) == ok
