@@ -147,7 +147,8 @@ struct Ensemble::Impl {
 
   mutable std::mutex session_mu;
   std::unordered_map<int64_t, SessionPtr> sessions;
-  std::unordered_set<int64_t> dead_sessions;  // expired or closed: reconnect ⇒ expired
+  // no tombstone set needed: session ids are monotonic and never reused, so
+  // an expired/closed id simply misses `sessions` ⇒ expired handshake
   int64_t next_session = 0x100000;
 
   mutable std::mutex conns_mu;
@@ -1343,7 +1344,6 @@ struct Ensemble::Impl {
       if (it == sessions.end()) return;
       s = it->second;
       sessions.erase(it);
-      dead_sessions.insert(sid);
     }
     s->alive.store(false, std::memory_order_release);
     std::unordered_set<std::string> eph;
