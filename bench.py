@@ -127,6 +127,14 @@ def main():
                          "matching BASELINE configs 1 vs 2/3)")
     args = ap.parse_args()
 
+    # The driver contract is ONE JSON line on stdout from rank 0 — but gloo
+    # writes "[Gloo] Rank ... connected ..." banners straight to fd 1 via
+    # std::cout (TORCH_CPP_LOG_LEVEL does not cover them). Route ALL
+    # incidental stdout to stderr at the fd level and keep the real stdout
+    # for the single result line.
+    real_stdout = os.dup(1)
+    os.dup2(2, 1)
+
     import registrar_amd as ra
 
     rank = int(os.environ.get("RANK", "0"))
@@ -329,7 +337,7 @@ def main():
                 "envelope_check_heartbeat_under_ms": 3000,
             },
         }
-        print(json.dumps(result), flush=True)
+        os.write(real_stdout, (json.dumps(result) + "\n").encode())
 
     client.close()
     if dist is not None:
