@@ -179,3 +179,23 @@ def test_session_timeout_expires_ephemerals(ensemble):
     ensemble.restart_server(0)
     assert wait_for(lambda: c.state() == "expired", timeout=10)
     c.close()
+
+
+def test_initial_connect_skips_dead_first_server(ensemble):
+    """Connect-string failover at START: the first server in the list is
+    unreachable; the client must rotate to the live one without burning a
+    full backoff cycle per attempt forever (lib/zk.js failover semantics)."""
+    from conftest import free_port
+
+    dead = ("127.0.0.1", free_port())
+    live_host, live_port = ensemble.connect_string().rsplit(":", 1)
+    c = ra.ZkClient(servers=[dead, (live_host, int(live_port))],
+                    session_timeout_ms=10000, randomize_start=False,
+                    connect_initial_delay_ms=50, connect_max_delay_ms=100)
+    c.start()
+    try:
+        assert c.wait_connected(15000), "failover to the live server failed"
+        rc, _ = c.create("/fo", b"", True)
+        assert rc == ra.ZOK
+    finally:
+        c.close()
