@@ -377,7 +377,7 @@ RegisterResult register_prepared(zk::ZkClient& client, PreparedRegistration& pre
     }
   }
 
-    result.znodes = prep.nodes;
+  result.znodes = prep.nodes;
 
   // 5) registerService: persistent put of the service record at $path itself,
   //    appended to the heartbeat node list (lib/register.js:45-75)
