@@ -202,3 +202,30 @@ def test_verbose_flag_steps_log_level(ensemble, daemon_bin):
     os.unlink(cfg_path)
     assert any(r["level"] == 20 for r in recs), "one -v must enable debug (level 20) records"
     assert not any(r["level"] == 10 for r in recs), "one -v must not reach trace"
+
+
+def test_zkensembled_binary_quickstart():
+    """The README quickstart path: spawn the standalone synthetic ensemble
+    binary, parse its JSON ports line, speak ZK to it, shut it down clean."""
+    bin_path = os.path.join(REPO_ROOT, "bin", "zkensembled")
+    assert os.path.exists(bin_path), "run `make daemon` first"
+    proc = subprocess.Popen([bin_path, "-n", "3", "--tick-ms", "100"],
+                            stdout=subprocess.PIPE, stderr=subprocess.DEVNULL, text=True)
+    try:
+        line = proc.stdout.readline()
+        info = json.loads(line)
+        assert len(info["ports"]) == 3
+        servers = [("127.0.0.1", p) for p in info["ports"]]
+        import registrar_amd as ra
+
+        c = ra.ZkClient(servers=servers, session_timeout_ms=10000)
+        c.start()
+        assert c.wait_connected(15000)
+        rc, _ = c.create("/standalone", b"x", True)
+        assert rc == ra.ZOK
+        rc, data, _ = c.get("/standalone")
+        assert rc == ra.ZOK and data == b"x"
+        c.close()
+    finally:
+        proc.send_signal(signal.SIGTERM)
+        assert proc.wait(timeout=10) == 0
