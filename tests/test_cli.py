@@ -130,3 +130,42 @@ def test_check_rejects_bad_stdout_match(tmp_path):
     p2.write_text(json.dumps(bad_flag))
     r = run_cli("check", "-f", str(p2))
     assert r.returncode == 1 and "unsupported flag" in r.stderr
+
+
+def test_binder_subcommand(ensemble, client):
+    """README quickstart: `python -m registrar_amd binder` answers DNS from
+    the registration tree; SIGTERM exits 0."""
+    import signal as _signal
+    import socket
+    import struct
+    import subprocess
+    import sys
+
+    import registrar_amd as ra
+
+    reg = {"domain": "cli.binder.test", "type": "host", "adminIp": "10.5.0.9",
+           "hostname": "cb0", "settleMs": 0}
+    rc, err, _ = ra.register_node(client, json.dumps(reg))
+    assert rc == ra.ZOK
+
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "registrar_amd", "binder",
+         "--servers", ensemble.connect_string(), "--port", "0"],
+        stdout=subprocess.PIPE, stderr=subprocess.DEVNULL, text=True, cwd=REPO_ROOT)
+    try:
+        line = proc.stdout.readline()
+        host, port = json.loads(line)["dns"].rsplit(":", 1)
+        q = struct.pack(">HHHHHH", 0x31ca, 0x0100, 1, 0, 0, 0)
+        for label in "cli.binder.test".split("."):
+            q += bytes([len(label)]) + label.encode()
+        q += b"\x00" + struct.pack(">HH", 1, 1)
+        s = socket.socket(socket.AF_INET, socket.SOCK_DGRAM)
+        s.settimeout(5)
+        s.sendto(q, (host, int(port)))
+        buf, _ = s.recvfrom(512)
+        s.close()
+        assert struct.unpack(">H", buf[6:8])[0] == 1  # one A answer
+        assert socket.inet_ntoa(buf[-4:]) == "10.5.0.9"
+    finally:
+        proc.send_signal(_signal.SIGTERM)
+        assert proc.wait(timeout=10) == 0
