@@ -169,3 +169,25 @@ def test_binder_subcommand(ensemble, client):
     finally:
         proc.send_signal(_signal.SIGTERM)
         assert proc.wait(timeout=10) == 0
+
+
+def test_consumer_example(ensemble, client):
+    """examples/consumer.py lists live instances with their xGMI rank — the
+    documented Binder-style consumer pattern."""
+    import registrar_amd as ra
+
+    reg = {"domain": "svc.consumer.test", "type": "host", "adminIp": "10.6.0.2",
+           "hostname": "cx0", "settleMs": 0,
+           "gpu": {"index": 3, "xgmiRank": 5, "uuid": "GPU-test"},
+           "service": {"type": "service",
+                       "service": {"srvce": "_infer", "proto": "_tcp", "port": 8000, "ttl": 30}}}
+    rc, err, _ = ra.register_node(client, json.dumps(reg))
+    assert rc == ra.ZOK
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO_ROOT, "examples", "consumer.py"),
+         "--servers", ensemble.connect_string(), "svc.consumer.test"],
+        capture_output=True, text=True, timeout=60, cwd=REPO_ROOT)
+    assert r.returncode == 0, r.stderr
+    assert "1 live instance(s)" in r.stdout
+    assert "10.6.0.2" in r.stdout and "xgmiRank=5" in r.stdout
+    assert "SRV _infer._tcp.svc.consumer.test port=8000" in r.stdout
