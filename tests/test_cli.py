@@ -191,3 +191,15 @@ def test_consumer_example(ensemble, client):
     assert "1 live instance(s)" in r.stdout
     assert "10.6.0.2" in r.stdout and "xgmiRank=5" in r.stdout
     assert "SRV _infer._tcp.svc.consumer.test port=8000" in r.stdout
+
+
+def test_shipped_sample_configs_are_valid():
+    """Every etc/*.json sample must pass `registrard check`-level validation
+    (schema drift in shipped samples is a doc bug users hit first)."""
+    import glob
+
+    samples = sorted(glob.glob(os.path.join(REPO_ROOT, "etc", "*.json")))
+    assert samples, "no sample configs found"
+    for path in samples:
+        r = run_cli("check", "-f", path)
+        assert r.returncode == 0, "%s failed check: %s" % (path, r.stderr)
