@@ -8,6 +8,7 @@ result is cross-checked against the model. Watch semantics are modeled too
 consumption, NodeDeleted union-delivery to data+child watchers, child events
 on the parent, and watch death across session expiry."""
 import json
+import os
 import time
 from collections import Counter
 
@@ -297,8 +298,8 @@ class EnsembleMachine(RuleBasedStateMachine):
 
 TestEnsembleModel = EnsembleMachine.TestCase
 TestEnsembleModel.settings = settings(
-    max_examples=40,
-    stateful_step_count=40,
+    max_examples=int(os.environ.get("MODEL_EXAMPLES", "60")),
+    stateful_step_count=50,
     deadline=None,
     suppress_health_check=[HealthCheck.too_slow],
 )
